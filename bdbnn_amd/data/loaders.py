@@ -1,0 +1,208 @@
+"""Data pipeline (ref:loader.py), rebuilt without torchvision.
+
+Same entry points and defaults as the reference
+(dataloader_cifar10/cifar100/imagenet, ref:loader.py:7,31,52), plus:
+
+* a real ``DistributedSampler`` with ``set_epoch`` wired in — the
+  reference's distributed branch is broken and each DDP rank iterated
+  the full dataset (ref:loader.py:67 + ref:train.py:372-373);
+* ``synthetic=True`` / ``dataloader_synthetic`` — ImageNet/CIFAR-shaped
+  random data with deterministic per-index generation (BASELINE.json
+  benchmarks run on synthetic data; this offline image has no datasets);
+* pure-numpy CIFAR reader (python-pickle batches) and a minimal
+  ImageFolder reader gated on PIL availability.
+
+Augmentations (random crop + horizontal flip + normalize) are
+implemented as tensor ops inside the datasets.
+"""
+
+import os
+import pickle
+
+import numpy as np
+import torch
+from torch.utils.data import DataLoader, Dataset
+from torch.utils.data.distributed import DistributedSampler
+
+_CIFAR_MEAN = (0.4914, 0.4822, 0.4465)
+_CIFAR_STD = (0.2023, 0.1994, 0.2010)
+_IMAGENET_MEAN = (0.485, 0.456, 0.406)
+_IMAGENET_STD = (0.229, 0.224, 0.225)
+
+
+class SyntheticImageDataset(Dataset):
+    """Deterministic random images + labels of a given shape.
+
+    Each index generates its sample from a per-index torch.Generator so
+    the data is reproducible and worker-independent.
+    """
+
+    def __init__(self, length, image_shape=(3, 224, 224), num_classes=1000,
+                 seed=0, dtype=torch.float32):
+        self.length = length
+        self.image_shape = tuple(image_shape)
+        self.num_classes = num_classes
+        self.seed = seed
+        self.dtype = dtype
+
+    def __len__(self):
+        return self.length
+
+    def __getitem__(self, idx):
+        g = torch.Generator().manual_seed(self.seed * 1000003 + idx)
+        img = torch.randn(self.image_shape, generator=g, dtype=self.dtype)
+        label = int(torch.randint(self.num_classes, (1,), generator=g))
+        return img, label
+
+
+def _normalize(img, mean, std):
+    mean = torch.tensor(mean, dtype=img.dtype).view(3, 1, 1)
+    std = torch.tensor(std, dtype=img.dtype).view(3, 1, 1)
+    return (img - mean) / std
+
+
+class CIFARBase(Dataset):
+    """CIFAR from the python-pickle batch files (no torchvision)."""
+
+    mean = _CIFAR_MEAN
+    std = _CIFAR_STD
+
+    def __init__(self, root, train=True, augment=True):
+        self.train = train
+        self.augment = augment and train
+        data, labels = [], []
+        for fname in self._files(train):
+            path = os.path.join(root, self._subdir(), fname)
+            with open(path, "rb") as f:
+                d = pickle.load(f, encoding="latin1")
+            data.append(d["data"])
+            labels.extend(d.get("labels", d.get("fine_labels")))
+        self.data = np.concatenate(data).reshape(-1, 3, 32, 32)
+        self.labels = np.asarray(labels, dtype=np.int64)
+
+    def __len__(self):
+        return len(self.labels)
+
+    def __getitem__(self, idx):
+        img = torch.from_numpy(self.data[idx].astype(np.float32) / 255.0)
+        if self.augment:
+            # random crop with pad 4 + horizontal flip (ref:loader.py:10-12)
+            img = torch.nn.functional.pad(img, (4, 4, 4, 4))
+            i = torch.randint(0, 9, (2,))
+            img = img[:, i[0]:i[0] + 32, i[1]:i[1] + 32]
+            if torch.rand(()) < 0.5:
+                img = img.flip(-1)
+        img = _normalize(img, self.mean, self.std)
+        return img, int(self.labels[idx])
+
+
+class CIFAR10Dataset(CIFARBase):
+    def _subdir(self):
+        return "cifar-10-batches-py"
+
+    def _files(self, train):
+        return ([f"data_batch_{i}" for i in range(1, 6)] if train
+                else ["test_batch"])
+
+
+class CIFAR100Dataset(CIFARBase):
+    def _subdir(self):
+        return "cifar-100-python"
+
+    def _files(self, train):
+        return ["train"] if train else ["test"]
+
+
+class ImageFolderDataset(Dataset):
+    """Minimal ImageNet-style folder reader (requires PIL)."""
+
+    def __init__(self, root, train=True):
+        try:
+            from PIL import Image  # noqa: F401
+        except ImportError as e:
+            raise RuntimeError(
+                "ImageFolderDataset needs PIL, which is not installed in "
+                "this image; use synthetic=True") from e
+        self.root = root
+        self.train = train
+        classes = sorted(d for d in os.listdir(root)
+                         if os.path.isdir(os.path.join(root, d)))
+        self.samples = []
+        for ci, c in enumerate(classes):
+            cdir = os.path.join(root, c)
+            for fn in sorted(os.listdir(cdir)):
+                self.samples.append((os.path.join(cdir, fn), ci))
+
+    def __len__(self):
+        return len(self.samples)
+
+    def __getitem__(self, idx):
+        from PIL import Image
+        path, label = self.samples[idx]
+        img = Image.open(path).convert("RGB")
+        if self.train:
+            img = img.resize((224, 224))
+            t = torch.from_numpy(np.asarray(img, np.float32) / 255.0).permute(2, 0, 1)
+            if torch.rand(()) < 0.5:
+                t = t.flip(-1)
+        else:
+            img = img.resize((256, 256))
+            t = torch.from_numpy(np.asarray(img, np.float32) / 255.0).permute(2, 0, 1)
+            t = t[:, 16:240, 16:240]
+        return _normalize(t, _IMAGENET_MEAN, _IMAGENET_STD), label
+
+
+def _make_loader(dataset, batch_size, shuffle, workers, distributed,
+                 pin_memory=True, drop_last=False):
+    sampler = None
+    if distributed:
+        sampler = DistributedSampler(dataset, shuffle=shuffle)
+        shuffle = False
+    return DataLoader(dataset, batch_size=batch_size, shuffle=shuffle,
+                      num_workers=workers, pin_memory=pin_memory,
+                      sampler=sampler, drop_last=drop_last,
+                      persistent_workers=workers > 0)
+
+
+def dataloader_cifar10(split="train", batch_size=128, data_path=None,
+                       distributed=False, workers=4, synthetic=False):
+    train = split == "train"
+    if synthetic or data_path is None or not os.path.isdir(
+            os.path.join(data_path or "", "cifar-10-batches-py")):
+        ds = SyntheticImageDataset(50000 if train else 10000, (3, 32, 32), 10)
+    else:
+        ds = CIFAR10Dataset(data_path, train=train)
+    return _make_loader(ds, batch_size, train, workers, distributed)
+
+
+def dataloader_cifar100(split="train", batch_size=128, data_path=None,
+                        distributed=False, workers=4, synthetic=False):
+    train = split == "train"
+    if synthetic or data_path is None or not os.path.isdir(
+            os.path.join(data_path or "", "cifar-100-python")):
+        ds = SyntheticImageDataset(50000 if train else 10000, (3, 32, 32), 100)
+    else:
+        ds = CIFAR100Dataset(data_path, train=train)
+    return _make_loader(ds, batch_size, train, workers, distributed)
+
+
+def dataloader_imagenet(split="train", batch_size=128, data_path="./",
+                        distributed=False, workers=8, synthetic=False,
+                        synthetic_len=100000):
+    train = split == "train"
+    sub = os.path.join(data_path, "train" if train else "val")
+    if synthetic or not os.path.isdir(sub):
+        ds = SyntheticImageDataset(synthetic_len if train else 10000,
+                                   (3, 224, 224), 1000)
+    else:
+        ds = ImageFolderDataset(sub, train=train)
+    return _make_loader(ds, batch_size, train, workers, distributed,
+                        drop_last=train)
+
+
+def dataloader_synthetic(batch_size, image_shape=(3, 224, 224),
+                         num_classes=1000, length=100000, workers=4,
+                         distributed=False):
+    ds = SyntheticImageDataset(length, image_shape, num_classes)
+    return _make_loader(ds, batch_size, True, workers, distributed,
+                        drop_last=True)

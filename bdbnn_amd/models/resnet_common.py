@@ -1,0 +1,238 @@
+"""ResNet skeletons (real-valued and binarized) with torchvision-compatible
+module naming.
+
+The weight-space KD (ops/kd.py, ref:utils/KD_loss.py:59-64) matches
+teacher and student conv modules BY NAME and needs identical weight
+shapes, so both the fp32 teacher and the 1-bit student mirror the
+torchvision ResNet layout and names exactly:
+conv1 / bn1 / layer{1..4}.{i}.conv1|bn1|conv2|bn2|downsample.0|downsample.1
+/ avgpool / fc.  (torchvision itself is not installed in this image;
+these are from-scratch implementations.)
+
+Binarized blocks follow the Bi-Real/ReActNet recipe: every binary conv
+carries its own residual connection and a PReLU-family activation;
+first conv and final fc stay real (ref contract).
+ResNet-18 has 19 convs besides the stem (16 block + 3 downsample) —
+matching the 19-entry per-layer kurtosis target lists
+(ref:train.py:467-470).
+"""
+
+import torch
+import torch.nn as nn
+
+from ..ops.binary_conv import (
+    HardBinaryConv,
+    HardBinaryConv_react,
+    HardBinaryConv_cifar,
+)
+from ..ops.binarize import LearnableBias
+
+
+class RPReLU(nn.Module):
+    """ReActNet RPReLU: shift -> PReLU -> shift, per channel."""
+
+    def __init__(self, channels):
+        super().__init__()
+        self.move1 = LearnableBias(channels)
+        self.prelu = nn.PReLU(channels)
+        self.move2 = LearnableBias(channels)
+
+    def forward(self, x):
+        return self.move2(self.prelu(self.move1(x)))
+
+
+def _make_act(kind, channels):
+    if kind == "relu":
+        return nn.ReLU(inplace=True)
+    if kind == "prelu":
+        return nn.PReLU(channels)
+    if kind == "rprelu":
+        return RPReLU(channels)
+    raise ValueError(kind)
+
+
+class BasicBlock(nn.Module):
+    """Real-valued torchvision-style BasicBlock."""
+
+    expansion = 1
+
+    def __init__(self, inplanes, planes, stride=1, downsample=None):
+        super().__init__()
+        self.conv1 = nn.Conv2d(inplanes, planes, 3, stride, 1, bias=False)
+        self.bn1 = nn.BatchNorm2d(planes)
+        self.relu = nn.ReLU(inplace=True)
+        self.conv2 = nn.Conv2d(planes, planes, 3, 1, 1, bias=False)
+        self.bn2 = nn.BatchNorm2d(planes)
+        self.downsample = downsample
+        self.stride = stride
+
+    def forward(self, x):
+        identity = self.downsample(x) if self.downsample is not None else x
+        out = self.relu(self.bn1(self.conv1(x)))
+        out = self.bn2(self.conv2(out))
+        out = self.relu(out + identity)
+        return out
+
+
+class BiBasicBlock(nn.Module):
+    """Binarized BasicBlock: per-conv residual (Bi-Real) + PReLU/RPReLU.
+
+    Module names (conv1/bn1/conv2/bn2/downsample.0/.1) match the real
+    block so weight-KD pairs by name with identical shapes.
+    """
+
+    expansion = 1
+
+    def __init__(self, inplanes, planes, stride=1, downsample=None,
+                 conv_cls=HardBinaryConv, act="prelu"):
+        super().__init__()
+        self.conv1 = conv_cls(inplanes, planes, 3, stride, 1)
+        self.bn1 = nn.BatchNorm2d(planes)
+        self.conv2 = conv_cls(planes, planes, 3, 1, 1)
+        self.bn2 = nn.BatchNorm2d(planes)
+        self.downsample = downsample
+        self.stride = stride
+        self.act1 = _make_act(act, planes)
+        self.act2 = _make_act(act, planes)
+
+    def forward(self, x):
+        identity = self.downsample(x) if self.downsample is not None else x
+        out = self.bn1(self.conv1(x))
+        out = self.act1(out + identity)
+        out2 = self.bn2(self.conv2(out))
+        out = self.act2(out2 + out)
+        return out
+
+
+class ResNet(nn.Module):
+    """Shared trunk.  stem='imagenet' (7x7/2 + maxpool) or 'cifar' (3x3/1)."""
+
+    def __init__(self, block_fn, layers, num_classes=1000, stem="imagenet",
+                 width=64, binary=False, conv_cls=None, act="prelu"):
+        super().__init__()
+        self.binary = binary
+        self.inplanes = width
+        if stem == "imagenet":
+            self.conv1 = nn.Conv2d(3, width, 7, 2, 3, bias=False)
+        else:
+            self.conv1 = nn.Conv2d(3, width, 3, 1, 1, bias=False)
+        self.bn1 = nn.BatchNorm2d(width)
+        self.relu = nn.ReLU(inplace=True)
+        self.maxpool = (nn.MaxPool2d(3, 2, 1) if stem == "imagenet"
+                        else nn.Identity())
+        stages = [width * (2 ** i) for i in range(len(layers))]
+        strides = [1] + [2] * (len(layers) - 1)
+        for i, (planes, blocks, stride) in enumerate(zip(stages, layers, strides), 1):
+            setattr(self, f"layer{i}",
+                    self._make_layer(block_fn, planes, blocks, stride,
+                                     conv_cls=conv_cls, act=act))
+        self.avgpool = nn.AdaptiveAvgPool2d((1, 1))
+        self.fc = nn.Linear(stages[-1], num_classes)
+
+        for m in self.modules():
+            if isinstance(m, nn.Conv2d):
+                nn.init.kaiming_normal_(m.weight, mode="fan_out", nonlinearity="relu")
+            elif isinstance(m, nn.BatchNorm2d):
+                nn.init.constant_(m.weight, 1)
+                nn.init.constant_(m.bias, 0)
+
+    def _make_layer(self, block_fn, planes, blocks, stride, conv_cls, act):
+        downsample = None
+        if stride != 1 or self.inplanes != planes:
+            if self.binary:
+                ds_conv = conv_cls(self.inplanes, planes, 1, stride, 0)
+            else:
+                ds_conv = nn.Conv2d(self.inplanes, planes, 1, stride, bias=False)
+            downsample = nn.Sequential(ds_conv, nn.BatchNorm2d(planes))
+        layers = []
+        kw = dict(conv_cls=conv_cls, act=act) if self.binary else {}
+        layers.append(block_fn(self.inplanes, planes, stride, downsample, **kw))
+        self.inplanes = planes
+        for _ in range(1, blocks):
+            layers.append(block_fn(self.inplanes, planes, **kw))
+        return nn.Sequential(*layers)
+
+    def forward(self, x):
+        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.layer1(x)
+        x = self.layer2(x)
+        x = self.layer3(x)
+        x = self.layer4(x) if hasattr(self, "layer4") else x
+        x = self.avgpool(x)
+        x = torch.flatten(x, 1)
+        return self.fc(x)
+
+
+def _resnet(layers, binary, conv_cls=None, act="prelu", **kw):
+    if binary:
+        return ResNet(BiBasicBlock, layers, binary=True,
+                      conv_cls=conv_cls, act=act, **kw)
+    return ResNet(BasicBlock, layers, binary=False, **kw)
+
+
+# ---------------- ImageNet constructors ----------------
+
+def resnet18_real(pretrained=False, num_classes=1000, stem="imagenet"):
+    return _resnet([2, 2, 2, 2], binary=False, num_classes=num_classes,
+                   stem=stem)
+
+
+def resnet34_real(pretrained=False, num_classes=1000, stem="imagenet"):
+    return _resnet([3, 4, 6, 3], binary=False, num_classes=num_classes,
+                   stem=stem)
+
+
+def resnet18_bi(pretrained=False, num_classes=1000, conv_cls=HardBinaryConv,
+                act="prelu", stem="imagenet"):
+    return _resnet([2, 2, 2, 2], binary=True, conv_cls=conv_cls, act=act,
+                   num_classes=num_classes, stem=stem)
+
+
+def resnet34_bi(pretrained=False, num_classes=1000, conv_cls=HardBinaryConv,
+                act="prelu", stem="imagenet"):
+    return _resnet([3, 4, 6, 3], binary=True, conv_cls=conv_cls, act=act,
+                   num_classes=num_classes, stem=stem)
+
+
+# ---------------- CIFAR constructors ----------------
+
+class CifarResNet(nn.Module):
+    """3-stage CIFAR ResNet (16/32/64 channels), real or binary."""
+
+    def __init__(self, block_fn, n_per_stage, num_classes=10, binary=False,
+                 conv_cls=None, act="prelu"):
+        super().__init__()
+        self.binary = binary
+        self.inplanes = 16
+        self.conv1 = nn.Conv2d(3, 16, 3, 1, 1, bias=False)
+        self.bn1 = nn.BatchNorm2d(16)
+        self.relu = nn.ReLU(inplace=True)
+        kw = dict(conv_cls=conv_cls, act=act) if binary else {}
+        self.layer1 = self._make_layer(block_fn, 16, n_per_stage, 1, kw)
+        self.layer2 = self._make_layer(block_fn, 32, n_per_stage, 2, kw)
+        self.layer3 = self._make_layer(block_fn, 64, n_per_stage, 2, kw)
+        self.avgpool = nn.AdaptiveAvgPool2d((1, 1))
+        self.fc = nn.Linear(64, num_classes)
+
+    def _make_layer(self, block_fn, planes, blocks, stride, kw):
+        downsample = None
+        if stride != 1 or self.inplanes != planes:
+            if self.binary:
+                ds_conv = kw["conv_cls"](self.inplanes, planes, 1, stride, 0)
+            else:
+                ds_conv = nn.Conv2d(self.inplanes, planes, 1, stride, bias=False)
+            downsample = nn.Sequential(ds_conv, nn.BatchNorm2d(planes))
+        layers = [block_fn(self.inplanes, planes, stride, downsample, **kw)]
+        self.inplanes = planes
+        for _ in range(1, blocks):
+            layers.append(block_fn(self.inplanes, planes, **kw))
+        return nn.Sequential(*layers)
+
+    def forward(self, x):
+        x = self.relu(self.bn1(self.conv1(x)))
+        x = self.layer1(x)
+        x = self.layer2(x)
+        x = self.layer3(x)
+        x = self.avgpool(x)
+        x = torch.flatten(x, 1)
+        return self.fc(x)
