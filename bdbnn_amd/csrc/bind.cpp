@@ -1,0 +1,319 @@
+// Python bindings for the bdbnn_amd gfx950 kernels (pack / xnor conv /
+// kurtosis / weight-KD / fused optimizers).  Pure HIP underneath — no CUDA
+// compatibility layer; this file only marshals ATen tensors.
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <vector>
+
+// ---- mirrors of csrc/common.h (kept in sync) ----
+constexpr int BDBNN_MAX_TENSORS = 64;
+struct TensorListArg {
+  const float* ptr[BDBNN_MAX_TENSORS];
+  int64_t numel[BDBNN_MAX_TENSORS];
+  int n;
+};
+struct PtrList { float* ptr[BDBNN_MAX_TENSORS]; };
+
+extern "C" {
+void bdbnn_sign_pack(const void*, uint32_t*, int64_t, int, int, bool,
+                     hipStream_t);
+void bdbnn_binsign_decode(const void*, void*, int64_t, bool, bool,
+                          hipStream_t);
+void bdbnn_ste_mask_mul(const void*, const void*, void*, int64_t, bool, bool,
+                        int, float, float, hipStream_t);
+void bdbnn_weight_pack(const float*, uint32_t*, float*, float*, int, int,
+                       int, int, int, hipStream_t);
+void bdbnn_xnor_conv_fwd(const uint32_t*, const uint32_t*, const float*,
+                         const float*, void*, bool, int, int, int, int, int,
+                         int, int, int, int, int, int, hipStream_t);
+void bdbnn_kurtosis_fwd(const TensorListArg*, const int*, const int64_t*,
+                        int, double*, const float*, float*, float*, float*,
+                        hipStream_t);
+void bdbnn_kurtosis_bwd(const TensorListArg*, const PtrList*, const int*,
+                        const int64_t*, int, const float*, const float*,
+                        float, hipStream_t);
+void bdbnn_weight_kd_fwd(const TensorListArg*, const PtrList*, const PtrList*,
+                         const int*, const int64_t*, int, double*,
+                         hipStream_t);
+void bdbnn_weight_kd_bwd(const TensorListArg*, const PtrList*, const PtrList*,
+                         const int*, const int64_t*, int, float, hipStream_t);
+void bdbnn_fused_sgd(const TensorListArg*, const PtrList*, const PtrList*,
+                     const PtrList*, const int*, const int64_t*, int, float,
+                     float, float, hipStream_t);
+void bdbnn_fused_adam(const TensorListArg*, const PtrList*, const PtrList*,
+                      const PtrList*, const PtrList*, const int*,
+                      const int64_t*, int, float, float, float, float, float,
+                      float, float, hipStream_t);
+}
+
+namespace {
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+bool is_bf16(const at::Tensor& t) { return t.scalar_type() == at::kBFloat16; }
+
+// Build the multi-tensor block schedule (block -> tensor, offset) on the
+// device.  chunk_elems must match the kernels' CHUNK_ELEMS.
+struct Schedule {
+  at::Tensor bt, bo;   // int32 / int64 device tensors
+  int n_blocks;
+};
+
+Schedule build_schedule(const std::vector<at::Tensor>& ts,
+                        int64_t chunk_elems, const at::Device& dev) {
+  std::vector<int> bt;
+  std::vector<int64_t> bo;
+  for (size_t l = 0; l < ts.size(); ++l) {
+    int64_t n = ts[l].numel();
+    for (int64_t off = 0; off < n; off += chunk_elems) {
+      bt.push_back((int)l);
+      bo.push_back(off);
+    }
+  }
+  auto opts = at::TensorOptions().device(dev);
+  Schedule s;
+  s.n_blocks = (int)bt.size();
+  s.bt = at::from_blob(bt.data(), {(int64_t)bt.size()},
+                       at::TensorOptions().dtype(at::kInt))
+             .to(dev, /*non_blocking=*/false);
+  s.bo = at::from_blob(bo.data(), {(int64_t)bo.size()},
+                       at::TensorOptions().dtype(at::kLong))
+             .to(dev, /*non_blocking=*/false);
+  return s;
+}
+
+TensorListArg make_meta(const std::vector<at::Tensor>& ts) {
+  TORCH_CHECK((int)ts.size() <= BDBNN_MAX_TENSORS,
+              "too many tensors for one fused call");
+  TensorListArg a;
+  a.n = (int)ts.size();
+  for (size_t i = 0; i < ts.size(); ++i) {
+    // dense layout in ANY memory format: the kernels are elementwise /
+    // permutation-invariant over physical memory
+    TORCH_CHECK((ts[i].is_contiguous() ||
+                 ts[i].is_contiguous(at::MemoryFormat::ChannelsLast)) &&
+                    ts[i].scalar_type() == at::kFloat,
+                "fused multi-tensor ops need dense fp32");
+    a.ptr[i] = ts[i].data_ptr<float>();
+    a.numel[i] = ts[i].numel();
+  }
+  return a;
+}
+
+PtrList make_ptrs(const std::vector<at::Tensor>& ts) {
+  PtrList p;
+  for (size_t i = 0; i < ts.size(); ++i) p.ptr[i] = ts[i].data_ptr<float>();
+  return p;
+}
+
+}  // namespace
+
+// ---------------- pack / quantizer ----------------
+
+at::Tensor sign_pack_nhwc(const at::Tensor& x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4, "sign_pack: 4-D CUDA tensor");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "sign_pack: channels_last input required");
+  int64_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int CW = (int)((C + 31) / 32);
+  auto out = at::empty({N, H, W, CW},
+                       x.options().dtype(at::kInt));
+  bdbnn_sign_pack(x.data_ptr(), (uint32_t*)out.data_ptr<int>(),
+                  N * H * W, (int)C, CW, is_bf16(x), cur_stream());
+  return out;
+}
+
+std::vector<at::Tensor> weight_pack(const at::Tensor& w) {
+  TORCH_CHECK(w.is_cuda() && w.dim() == 4, "weight_pack: 4-D CUDA tensor");
+  auto wf = w.contiguous().to(at::kFloat);
+  int K = (int)wf.size(0), C = (int)wf.size(1);
+  int KH = (int)wf.size(2), KW = (int)wf.size(3);
+  int CW = (C + 31) / 32;
+  auto wp = at::empty({K, KH, KW, CW}, wf.options().dtype(at::kInt));
+  auto alpha = at::empty({K}, wf.options());
+  auto stab = at::empty({K, KH * KW}, wf.options());
+  bdbnn_weight_pack(wf.data_ptr<float>(), (uint32_t*)wp.data_ptr<int>(),
+                    alpha.data_ptr<float>(), stab.data_ptr<float>(), K, C,
+                    KH, KW, CW, cur_stream());
+  return {wp, alpha, stab};
+}
+
+at::Tensor binsign_decode(const at::Tensor& x, bool out_bf16) {
+  TORCH_CHECK(x.is_cuda(), "binsign_decode: CUDA tensor");
+  auto fmt = x.dim() == 4 ? at::MemoryFormat::ChannelsLast
+                          : at::MemoryFormat::Contiguous;
+  auto xc = x.contiguous(fmt);
+  auto out = at::empty_like(xc, xc.options().dtype(
+      out_bf16 ? at::kBFloat16 : at::kFloat), fmt);
+  bdbnn_binsign_decode(xc.data_ptr(), out.data_ptr(), xc.numel(),
+                       is_bf16(xc), out_bf16, cur_stream());
+  return out;
+}
+
+at::Tensor ste_mask_mul(const at::Tensor& g, const at::Tensor& x, int mode,
+                        double t, double k) {
+  TORCH_CHECK(g.is_cuda() && x.is_cuda() && g.numel() == x.numel(),
+              "ste_mask_mul: matching CUDA tensors");
+  auto fmt = x.dim() == 4 ? at::MemoryFormat::ChannelsLast
+                          : at::MemoryFormat::Contiguous;
+  auto xc = x.contiguous(fmt);
+  auto gc = g.contiguous(fmt);
+  auto out = at::empty_like(xc, xc.options(), fmt);
+  bdbnn_ste_mask_mul(gc.data_ptr(), xc.data_ptr(), out.data_ptr(),
+                     xc.numel(), is_bf16(gc), is_bf16(xc), mode, (float)t,
+                     (float)k, cur_stream());
+  return out;
+}
+
+// ---------------- xnor conv ----------------
+
+at::Tensor xnor_conv_fwd(const at::Tensor& xp, const at::Tensor& wp,
+                         const at::Tensor& alpha, const at::Tensor& stab,
+                         int64_t C, int64_t stride, int64_t pad,
+                         bool out_bf16) {
+  TORCH_CHECK(xp.is_cuda() && xp.dim() == 4 && xp.scalar_type() == at::kInt,
+              "xnor_conv: packed activations int32 [N,H,W,CW]");
+  TORCH_CHECK(wp.dim() == 4, "xnor_conv: packed weights [K,KH,KW,CW]");
+  int N = (int)xp.size(0), H = (int)xp.size(1), W = (int)xp.size(2);
+  int K = (int)wp.size(0), KH = (int)wp.size(1), KW = (int)wp.size(2);
+  int Ho = (int)((H + 2 * pad - KH) / stride + 1);
+  int Wo = (int)((W + 2 * pad - KW) / stride + 1);
+  auto out = at::empty({N, K, Ho, Wo},
+                       xp.options().dtype(out_bf16 ? at::kBFloat16
+                                                   : at::kFloat),
+                       at::MemoryFormat::ChannelsLast);
+  bdbnn_xnor_conv_fwd((const uint32_t*)xp.data_ptr<int>(),
+                      (const uint32_t*)wp.data_ptr<int>(),
+                      alpha.data_ptr<float>(), stab.data_ptr<float>(),
+                      out.data_ptr(), out_bf16, N, H, W, (int)C, K, KH, KW,
+                      (int)stride, (int)pad, Ho, Wo, cur_stream());
+  return out;
+}
+
+// ---------------- kurtosis ----------------
+
+std::vector<at::Tensor> kurtosis_fwd(const std::vector<at::Tensor>& ws,
+                                     const at::Tensor& targets) {
+  auto meta = make_meta(ws);
+  auto dev = ws[0].device();
+  auto sched = build_schedule(ws, 256 * 1024, dev);
+  int L = (int)ws.size();
+  auto opts = ws[0].options();
+  auto mom = at::zeros({L, 4}, opts.dtype(at::kDouble));
+  auto stats = at::empty({L, 4}, opts);
+  auto losses = at::empty({L}, opts);
+  auto kurts = at::empty({L}, opts);
+  auto tgt = targets.to(dev, at::kFloat).contiguous();
+  bdbnn_kurtosis_fwd(&meta, sched.bt.data_ptr<int>(),
+                     sched.bo.data_ptr<int64_t>(), sched.n_blocks,
+                     mom.data_ptr<double>(), tgt.data_ptr<float>(),
+                     stats.data_ptr<float>(), losses.data_ptr<float>(),
+                     kurts.data_ptr<float>(), cur_stream());
+  return {losses, kurts, stats};
+}
+
+std::vector<at::Tensor> kurtosis_bwd(const std::vector<at::Tensor>& ws,
+                                     const at::Tensor& stats,
+                                     const at::Tensor& targets,
+                                     double gscale) {
+  auto meta = make_meta(ws);
+  auto dev = ws[0].device();
+  auto sched = build_schedule(ws, 256 * 1024, dev);
+  std::vector<at::Tensor> grads;
+  for (auto& w : ws)  // preserve_format: grad layout == weight layout
+    grads.push_back(at::empty_like(w, w.options(),
+                                   at::MemoryFormat::Preserve));
+  auto gp = make_ptrs(grads);
+  auto tgt = targets.to(dev, at::kFloat).contiguous();
+  auto st = stats.contiguous();
+  bdbnn_kurtosis_bwd(&meta, &gp, sched.bt.data_ptr<int>(),
+                     sched.bo.data_ptr<int64_t>(), sched.n_blocks,
+                     st.data_ptr<float>(), tgt.data_ptr<float>(),
+                     (float)gscale, cur_stream());
+  return grads;
+}
+
+// ---------------- weight KD ----------------
+
+at::Tensor weight_kd_fwd(const std::vector<at::Tensor>& ws,
+                         const std::vector<at::Tensor>& wt) {
+  auto meta = make_meta(ws);
+  auto sched = build_schedule(ws, 256 * 1024, ws[0].device());
+  auto s_ptr = make_ptrs(ws);
+  std::vector<at::Tensor> wt_c;
+  for (auto& t : wt) wt_c.push_back(t.contiguous());
+  auto t_ptr = make_ptrs(wt_c);
+  auto out = at::zeros({}, ws[0].options().dtype(at::kDouble));
+  bdbnn_weight_kd_fwd(&meta, &s_ptr, &t_ptr, sched.bt.data_ptr<int>(),
+                      sched.bo.data_ptr<int64_t>(), sched.n_blocks,
+                      out.data_ptr<double>(), cur_stream());
+  return out.to(at::kFloat);
+}
+
+std::vector<at::Tensor> weight_kd_bwd(const std::vector<at::Tensor>& wt,
+                                      double gscale) {
+  auto meta = make_meta(wt);
+  auto sched = build_schedule(wt, 256 * 1024, wt[0].device());
+  auto t_ptr = make_ptrs(wt);
+  std::vector<at::Tensor> grads;
+  for (auto& t : wt)
+    grads.push_back(at::empty_like(t, t.options(),
+                                   at::MemoryFormat::Preserve));
+  auto gp = make_ptrs(grads);
+  bdbnn_weight_kd_bwd(&meta, &t_ptr, &gp, sched.bt.data_ptr<int>(),
+                      sched.bo.data_ptr<int64_t>(), sched.n_blocks,
+                      (float)gscale, cur_stream());
+  return grads;
+}
+
+// ---------------- fused optimizers ----------------
+
+void fused_sgd(const std::vector<at::Tensor>& p,
+               const std::vector<at::Tensor>& g,
+               const std::vector<at::Tensor>& buf, double lr, double momentum,
+               double wd) {
+  auto meta = make_meta(p);
+  auto sched = build_schedule(p, 256 * 512, p[0].device());
+  auto pp = make_ptrs(p);
+  auto gg = make_ptrs(g);   // python aligns grad layout to the param's
+  auto bb = make_ptrs(buf);
+  bdbnn_fused_sgd(&meta, &pp, &gg, &bb, sched.bt.data_ptr<int>(),
+                  sched.bo.data_ptr<int64_t>(), sched.n_blocks, (float)lr,
+                  (float)momentum, (float)wd, cur_stream());
+}
+
+void fused_adam(const std::vector<at::Tensor>& p,
+                const std::vector<at::Tensor>& g,
+                const std::vector<at::Tensor>& m1,
+                const std::vector<at::Tensor>& m2, double lr, double beta1,
+                double beta2, double eps, double wd, double bc1, double bc2) {
+  auto meta = make_meta(p);
+  auto sched = build_schedule(p, 256 * 512, p[0].device());
+  auto pp = make_ptrs(p);
+  auto gg = make_ptrs(g);   // python aligns grad layout to the param's
+  auto mm1 = make_ptrs(m1);
+  auto mm2 = make_ptrs(m2);
+  bdbnn_fused_adam(&meta, &pp, &gg, &mm1, &mm2, sched.bt.data_ptr<int>(),
+                   sched.bo.data_ptr<int64_t>(), sched.n_blocks, (float)lr,
+                   (float)beta1, (float)beta2, (float)eps, (float)wd,
+                   (float)bc1, (float)bc2, cur_stream());
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("sign_pack_nhwc", &sign_pack_nhwc, "sign+bitpack NHWC activations");
+  m.def("weight_pack", &weight_pack,
+        "pack conv weights -> (bits, alpha, pad-correction table)");
+  m.def("binsign_decode", &binsign_decode, "elementwise +-1 decode");
+  m.def("ste_mask_mul", &ste_mask_mul, "quantizer backward mask multiply");
+  m.def("xnor_conv_fwd", &xnor_conv_fwd, "bit-packed XNOR+popcount conv");
+  m.def("kurtosis_fwd", &kurtosis_fwd, "fused multi-tensor kurtosis fwd");
+  m.def("kurtosis_bwd", &kurtosis_bwd, "fused multi-tensor kurtosis bwd");
+  m.def("weight_kd_fwd", &weight_kd_fwd, "fused weight-space KD fwd");
+  m.def("weight_kd_bwd", &weight_kd_bwd, "fused weight-space KD bwd");
+  m.def("fused_sgd", &fused_sgd, "fused multi-tensor SGD-momentum step");
+  m.def("fused_adam", &fused_adam, "fused multi-tensor Adam step");
+}

@@ -1,0 +1,191 @@
+#include "hip/hip_runtime.h"
+// K2 — quantizer kernels: sign+bitpack, +-1 decode, STE/EDE backward masks,
+// weight pack (+ per-channel alpha and the per-tap padding-correction table).
+//
+// Bit conventions (the framework-wide contract, see ops/binary_conv.py):
+//   activation pack: bit c of word (p, cw) = 1  iff  x[p, 32*cw+c] >= 0
+//   weight pack:     bit c = 1               iff  w < 0        (INVERTED)
+// so that a_word XOR w_word == XNOR(sign_a, sign_w) and the conv inner loop
+// is two VALU ops per 32 channels.  Unused tail bits (C % 32 != 0): a-pack
+// stores 0, w-pack stores 1 — every garbage bit then contributes exactly 1
+// to the popcount sum, a compile-known constant the epilogue subtracts.
+#include "common.h"
+
+// ---------------- activation sign+pack ----------------
+// x: NHWC-contiguous (channels_last torch tensor), P = N*H*W pixels,
+// out: uint32 [P][CW].  One thread per output word.
+template <typename T>
+__global__ void sign_pack_kernel(const T* __restrict__ x,
+                                 uint32_t* __restrict__ out,
+                                 int64_t n_words, int C, int CW) {
+  GRID_STRIDE(i, n_words) {
+    int cw = int(i % CW);
+    int64_t p = i / CW;
+    const T* px = x + p * C + cw * 32;
+    int nbits = min(32, C - cw * 32);
+    uint32_t bits = 0;
+#pragma unroll 8
+    for (int c = 0; c < nbits; ++c) {
+      float v;
+      if constexpr (sizeof(T) == 2) v = bf16_to_f32(((const uint16_t*)px)[c]);
+      else                          v = ((const float*)px)[c];
+      bits |= (v >= 0.f ? 1u : 0u) << c;
+    }
+    out[i] = bits;
+  }
+}
+
+extern "C" void bdbnn_sign_pack(const void* x, uint32_t* out, int64_t pixels,
+                                int C, int CW, bool bf16,
+                                hipStream_t stream) {
+  int64_t n_words = pixels * CW;
+  int block = 256;
+  int grid = (int)bd_min<int64_t>((n_words + block - 1) / block, 65535 * 8);
+  if (bf16)
+   hipLaunchKernelGGL(( sign_pack_kernel<uint16_t>), dim3(grid), dim3(block), 0, stream, 
+        (const uint16_t*)x, out, n_words, C, CW);
+  else
+   hipLaunchKernelGGL(( sign_pack_kernel<float>), dim3(grid), dim3(block), 0, stream, 
+        (const float*)x, out, n_words, C, CW);
+}
+
+// ---------------- +-1 decode (for the dense MFMA backward) ----------------
+template <typename TI, typename TO>
+__global__ void binsign_decode_kernel(const TI* __restrict__ x,
+                                      TO* __restrict__ y, int64_t n) {
+  GRID_STRIDE(i, n) {
+    float v;
+    if constexpr (sizeof(TI) == 2) v = bf16_to_f32(((const uint16_t*)x)[i]);
+    else                           v = ((const float*)x)[i];
+    float s = v >= 0.f ? 1.f : -1.f;
+    if constexpr (sizeof(TO) == 2) ((uint16_t*)y)[i] = f32_to_bf16(s);
+    else                           ((float*)y)[i] = s;
+  }
+}
+
+extern "C" void bdbnn_binsign_decode(const void* x, void* y, int64_t n,
+                                     bool in_bf16, bool out_bf16,
+                                     hipStream_t stream) {
+  int block = 256;
+  int grid = (int)bd_min<int64_t>((n + block - 1) / block, 65535 * 8);
+#define CASE(IB, OB, TI, TO)                                              \
+  if (in_bf16 == IB && out_bf16 == OB)                                    \
+   hipLaunchKernelGGL(( binsign_decode_kernel<TI, TO>), dim3(grid), dim3(block), 0, stream,             \
+        (const TI*)x, (TO*)y, n);
+  CASE(false, false, float, float)
+  CASE(false, true, float, uint16_t)
+  CASE(true, false, uint16_t, float)
+  CASE(true, true, uint16_t, uint16_t)
+#undef CASE
+}
+
+// ---------------- STE/EDE mask multiply (quantizer backward) ----------------
+// y = g * mask(x); mode 0 = clip-STE 1(|x|<=1), 1 = ReActNet polynomial,
+// 2 = EDE k*t*(1-tanh^2(t*x)).  Output dtype = x dtype.
+template <typename TG, typename TX>
+__global__ void ste_mask_mul_kernel(const TG* __restrict__ g,
+                                    const TX* __restrict__ x,
+                                    TX* __restrict__ y, int64_t n,
+                                    int mode, float t, float k) {
+  GRID_STRIDE(i, n) {
+    float gv, xv;
+    if constexpr (sizeof(TG) == 2) gv = bf16_to_f32(((const uint16_t*)g)[i]);
+    else                           gv = ((const float*)g)[i];
+    if constexpr (sizeof(TX) == 2) xv = bf16_to_f32(((const uint16_t*)x)[i]);
+    else                           xv = ((const float*)x)[i];
+    float m;
+    if (mode == 0) {
+      m = fabsf(xv) <= 1.f ? 1.f : 0.f;
+    } else if (mode == 1) {
+      m = (xv >= -1.f && xv < 0.f) ? 2.f + 2.f * xv
+        : (xv >= 0.f && xv < 1.f) ? 2.f - 2.f * xv : 0.f;
+    } else {
+      float th = tanhf(t * xv);
+      m = k * t * (1.f - th * th);
+    }
+    float out = gv * m;
+    if constexpr (sizeof(TX) == 2) ((uint16_t*)y)[i] = f32_to_bf16(out);
+    else                           ((float*)y)[i] = out;
+  }
+}
+
+extern "C" void bdbnn_ste_mask_mul(const void* g, const void* x, void* y,
+                                   int64_t n, bool g_bf16, bool x_bf16,
+                                   int mode, float t, float k,
+                                   hipStream_t stream) {
+  int block = 256;
+  int grid = (int)bd_min<int64_t>((n + block - 1) / block, 65535 * 8);
+#define CASE(GB, XB, TG, TX)                                              \
+  if (g_bf16 == GB && x_bf16 == XB)                                       \
+   hipLaunchKernelGGL(( ste_mask_mul_kernel<TG, TX>), dim3(grid), dim3(block), 0, stream,               \
+        (const TG*)g, (const TX*)x, (TX*)y, n, mode, t, k);
+  CASE(false, false, float, float)
+  CASE(false, true, float, uint16_t)
+  CASE(true, false, uint16_t, float)
+  CASE(true, true, uint16_t, uint16_t)
+#undef CASE
+}
+
+// ---------------- weight pack ----------------
+// w: fp32 contiguous [K][C][KH][KW].
+// wp: uint32 [K][KH][KW][CW]   (bit = 1 iff w < 0; garbage tail bits = 1)
+// alpha: fp32 [K]              (mean |w[k]|)
+// stab:  fp32 [K][T]           (T = KH*KW; S[k][t] = C - 2*popc_real(t))
+//                              the zero-pad epilogue correction table.
+__global__ void weight_alpha_kernel(const float* __restrict__ w,
+                                    float* __restrict__ alpha,
+                                    int K, int64_t per_k) {
+  __shared__ float red[256];
+  int k = blockIdx.x;
+  const float* wk = w + (int64_t)k * per_k;
+  float s = 0.f;
+  for (int64_t i = threadIdx.x; i < per_k; i += blockDim.x)
+    s += fabsf(wk[i]);
+  red[threadIdx.x] = s;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) alpha[k] = red[0] / float(per_k);
+}
+
+__global__ void weight_pack_kernel(const float* __restrict__ w,
+                                   uint32_t* __restrict__ wp,
+                                   float* __restrict__ stab,
+                                   int K, int C, int KH, int KW, int CW) {
+  // one thread per packed word (k, t, cw); w layout stride: w[k][c][kh][kw]
+  int T = KH * KW;
+  int64_t n_words = (int64_t)K * T * CW;
+  GRID_STRIDE(i, n_words) {
+    int cw = int(i % CW);
+    int t = int((i / CW) % T);
+    int k = int(i / CW / T);
+    int nbits = min(32, C - cw * 32);
+    const float* wk = w + ((int64_t)k * C) * T + t;  // + c*T steps
+    uint32_t bits = 0;
+    int pop_real = 0;  // popc of REAL (non-inverted) sign bits
+    for (int c = 0; c < nbits; ++c) {
+      float v = wk[(int64_t)(cw * 32 + c) * T];
+      if (v < 0.f) bits |= 1u << c;       // inverted convention
+      else ++pop_real;
+    }
+    // garbage tail bits = 1 (so XOR with a-pack's 0 counts exactly 1)
+    if (nbits < 32) bits |= ~((nbits >= 32) ? 0xffffffffu : ((1u << nbits) - 1u));
+    wp[i] = bits;
+    // S[k][t] = C - 2 * popc_real(tap): reduce across cw via atomic
+    atomicAdd(&stab[(int64_t)k * T + t], float(nbits - 2 * pop_real));
+  }
+}
+
+extern "C" void bdbnn_weight_pack(const float* w, uint32_t* wp, float* alpha,
+                                  float* stab, int K, int C, int KH, int KW,
+                                  int CW, hipStream_t stream) {
+  int64_t per_k = (int64_t)C * KH * KW;
+ hipLaunchKernelGGL(( weight_alpha_kernel), dim3(K), dim3(256), 0, stream, w, alpha, K, per_k);
+  int64_t n_words = (int64_t)K * KH * KW * CW;
+  int block = 256;
+  int grid = (int)bd_min<int64_t>((n_words + block - 1) / block, 65535 * 8);
+  hipMemsetAsync(stab, 0, sizeof(float) * K * KH * KW, stream);
+ hipLaunchKernelGGL(( weight_pack_kernel), dim3(grid), dim3(block), 0, stream, w, wp, stab, K, C, KH, KW, CW);
+}

@@ -64,11 +64,12 @@ class BinaryConvFunction(torch.autograd.Function):
         ctx.save_for_backward(x, w)
         if x.is_cuda:
             nat = _C.native_required()
-            xp = nat.sign_pack_nhwc(x)          # uint32 [N,H,W,C/32]
-            wp, alpha = nat.weight_pack(w)      # uint32 [K,kh,kw,C/32], fp32 [K]
+            xc = x.contiguous(memory_format=torch.channels_last)
+            xp = nat.sign_pack_nhwc(xc)           # int32 [N,H,W,C/32]
+            wp, alpha, stab = nat.weight_pack(w)  # bits, alpha[K], S[K][T]
             out = nat.xnor_conv_fwd(
-                xp, wp, alpha, x.shape[1], stride, padding,
-                x.shape[2], x.shape[3], 1 if x.dtype == torch.bfloat16 else 0)
+                xp, wp, alpha, stab, x.shape[1], stride, padding,
+                x.dtype == torch.bfloat16)
             return out
         xb = binsign(x)
         wb = weight_scale(w) * binsign(w)

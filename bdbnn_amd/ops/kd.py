@@ -65,6 +65,16 @@ def _kl_log_target_mean(w_s, w_t):
     return (torch.exp(w_t) * (w_t - w_s)).mean()
 
 
+def _match_layout(t, like):
+    """Return t with the same dense memory layout as `like`."""
+    if t.stride() == like.stride():
+        return t
+    if like.dim() == 4 and like.is_contiguous(
+            memory_format=torch.channels_last):
+        return t.contiguous(memory_format=torch.channels_last)
+    return t.contiguous()
+
+
 class DistributionLoss_layer(_loss._Loss):
     """Weight-space KD via per-step module walk (API parity, ref:KD_loss.py:46-67)."""
 
@@ -105,8 +115,10 @@ class _FusedWeightKD(torch.autograd.Function):
         wt = tensors[n_pairs:]
         if ws[0].is_cuda and _C.has_native():
             nat = _C.native_required()
-            out = nat.weight_kd_fwd([a.reshape(-1) for a in ws],
-                                    [b.reshape(-1) for b in wt])
+            # elementwise pairing happens over physical memory: align the
+            # teacher tensor's layout to the student's
+            wt = tuple(_match_layout(b, a) for a, b in zip(ws, wt))
+            out = nat.weight_kd_fwd(list(ws), list(wt))
             ctx.save_for_backward(*wt)
             ctx.native = True
             return out
@@ -123,8 +135,7 @@ class _FusedWeightKD(torch.autograd.Function):
         grads = []
         if ctx.native:
             nat = _C.native_required()
-            grads = nat.weight_kd_bwd([b.reshape(-1) for b in wt], float(g))
-            grads = [gr.view_as(b) for gr, b in zip(grads, wt)]
+            grads = list(nat.weight_kd_bwd(list(wt), float(g)))
         else:
             for b in wt:
                 grads.append(-g * torch.exp(b) / b.numel())

@@ -90,9 +90,9 @@ class _FusedKurtosis(torch.autograd.Function):
     def forward(ctx, targets, *tensors):
         if tensors[0].is_cuda:
             nat = _C.native_required()
-            flat = [t.reshape(-1) for t in tensors]
-            # stats[l] = (mu, sigma_unbiased, kurt, mean(z^3)) per tensor
-            losses, kurts, stats = nat.kurtosis_fwd(flat, targets)
+            # tensors passed in their native layout: the reductions are
+            # permutation-invariant over physical memory
+            losses, kurts, stats = nat.kurtosis_fwd(list(tensors), targets)
             ctx.save_for_backward(targets, stats, *tensors)
             ctx.native = True
             return losses.sum(), kurts
@@ -114,9 +114,8 @@ class _FusedKurtosis(torch.autograd.Function):
             targets, stats = ctx.saved_tensors[:2]
             tensors = ctx.saved_tensors[2:]
             nat = _C.native_required()
-            grads = nat.kurtosis_bwd([t.reshape(-1) for t in tensors],
-                                     stats, targets, float(g))
-            return (None, *[gr.view_as(t) for gr, t in zip(grads, tensors)])
+            grads = nat.kurtosis_bwd(list(tensors), stats, targets, float(g))
+            return (None, *grads)
         targets = ctx.saved_tensors[0]
         tensors = ctx.saved_tensors[1:]
         grads = []

@@ -16,6 +16,17 @@ import torch
 from .. import _C
 
 
+def _match_layout(t, like):
+    """Dense tensor t re-laid-out to match `like`'s strides (channels_last
+    conv params after .to(memory_format=channels_last))."""
+    if t.stride() == like.stride():
+        return t
+    if like.dim() == 4 and like.is_contiguous(
+            memory_format=torch.channels_last):
+        return t.contiguous(memory_format=torch.channels_last)
+    return t.contiguous()
+
+
 class FusedSGD(torch.optim.Optimizer):
     """SGD with momentum + weight decay; multi-tensor fused step on GPU."""
 
@@ -43,9 +54,8 @@ class FusedSGD(torch.optim.Optimizer):
             if not ps:
                 continue
             if ps[0].is_cuda and _C.has_native():
-                _C.native_required().fused_sgd(
-                    [p.reshape(-1) for p in ps], [g.reshape(-1) for g in gs],
-                    [b.reshape(-1) for b in bufs], lr, mom, wd)
+                gs = [_match_layout(g.float(), p) for p, g in zip(ps, gs)]
+                _C.native_required().fused_sgd(ps, gs, bufs, lr, mom, wd)
             else:
                 for p, g, b in zip(ps, gs, bufs):
                     if wd != 0:
@@ -92,10 +102,9 @@ class FusedAdam(torch.optim.Optimizer):
             bc1 = 1 - beta1 ** step_t
             bc2 = 1 - beta2 ** step_t
             if ps[0].is_cuda and _C.has_native():
+                gs = [_match_layout(g.float(), p) for p, g in zip(ps, gs)]
                 _C.native_required().fused_adam(
-                    [p.reshape(-1) for p in ps], [g.reshape(-1) for g in gs],
-                    [m.reshape(-1) for m in m1s], [v.reshape(-1) for v in m2s],
-                    lr, beta1, beta2, eps, wd, bc1, bc2)
+                    ps, gs, m1s, m2s, lr, beta1, beta2, eps, wd, bc1, bc2)
             else:
                 for p, g, m, v in zip(ps, gs, m1s, m2s):
                     if wd != 0:

@@ -1,0 +1,234 @@
+"""HIP kernel numerics on MI355X — every kernel vs the plain-PyTorch
+fp32 oracle defined by the CPU ops."""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+from bdbnn_amd import _C
+from bdbnn_amd.ops.binarize import binsign, weight_scale
+from bdbnn_amd.ops.binary_conv import BinaryConvFunction, HardBinaryConv
+from bdbnn_amd.ops.kurtosis import kurtosis_loss_fused
+from bdbnn_amd.ops.optim import FusedSGD, FusedAdam
+
+
+def _nat():
+    return _C.native_required()
+
+
+def _cl(x):
+    return x.contiguous(memory_format=torch.channels_last)
+
+
+# ---------------- pack / decode ----------------
+
+@pytest.mark.parametrize("C", [32, 64, 48, 96])
+def test_sign_pack_bits(C):
+    torch.manual_seed(0)
+    x = torch.randn(2, C, 5, 7, device="cuda")
+    xp = _nat().sign_pack_nhwc(_cl(x))
+    assert xp.shape == (2, 5, 7, (C + 31) // 32)
+    bits = xp.cpu().numpy().astype("uint32")
+    xs = x.permute(0, 2, 3, 1).cpu()  # NHWC
+    for n in range(2):
+        for h in range(5):
+            for w in range(7):
+                for c in range(C):
+                    want = 1 if xs[n, h, w, c].item() >= 0 else 0
+                    got = (int(bits[n, h, w, c // 32]) >> (c % 32)) & 1
+                    assert got == want
+                # garbage bits of the a-pack are 0
+                tail = C % 32
+                if tail:
+                    assert (int(bits[n, h, w, C // 32]) >> tail) == 0
+
+
+def test_binsign_decode():
+    x = torch.randn(3, 16, 4, 4, device="cuda")
+    y = _nat().binsign_decode(_cl(x), False)
+    assert torch.equal(y, _cl(binsign(x)))
+    yb = _nat().binsign_decode(_cl(x), True)
+    assert yb.dtype == torch.bfloat16
+    assert torch.equal(yb.float(), _cl(binsign(x)))
+
+
+def test_ste_mask_modes():
+    g = torch.randn(4, 8, 6, 6, device="cuda")
+    x = torch.randn(4, 8, 6, 6, device="cuda") * 2
+    out = _nat().ste_mask_mul(_cl(g), _cl(x), 0, 0.0, 0.0)
+    ref = _cl(g * (x.abs() <= 1).float())
+    assert torch.allclose(out, ref)
+    out = _nat().ste_mask_mul(_cl(g), _cl(x), 1, 0.0, 0.0)
+    neg = (x >= -1) & (x < 0)
+    pos = (x >= 0) & (x < 1)
+    m = torch.where(neg, 2 + 2 * x, torch.where(pos, 2 - 2 * x,
+                                                torch.zeros_like(x)))
+    assert torch.allclose(out, _cl(g * m), atol=1e-6)
+    out = _nat().ste_mask_mul(_cl(g), _cl(x), 2, 2.0, 1.5)
+    th = torch.tanh(2.0 * x)
+    assert torch.allclose(out, _cl(g * 1.5 * 2.0 * (1 - th * th)), atol=1e-5)
+
+
+# ---------------- xnor conv ----------------
+
+@pytest.mark.parametrize("shape", [
+    # (N, C, H, W, K, ksize, stride, pad)
+    (2, 64, 14, 14, 64, 3, 1, 1),
+    (2, 128, 9, 11, 96, 3, 1, 1),     # non-multiple-of-tile K, odd spatial
+    (1, 64, 8, 8, 128, 1, 2, 0),      # 1x1 stride-2 downsample
+    (2, 48, 10, 10, 64, 3, 2, 1),     # tail channel word + stride 2
+    (1, 512, 7, 7, 512, 3, 1, 1),     # deepest layer shape
+    (3, 96, 6, 6, 16, 3, 1, 1),       # small K
+])
+def test_xnor_conv_matches_fp32_conv(shape):
+    N, C, H, W, K, ks, stride, pad = shape
+    torch.manual_seed(1)
+    x = torch.randn(N, C, H, W, device="cuda")
+    w = torch.randn(K, C, ks, ks, device="cuda") * 0.5
+    nat = _nat()
+    xp = nat.sign_pack_nhwc(_cl(x))
+    wp, alpha, stab = nat.weight_pack(w)
+    out = nat.xnor_conv_fwd(xp, wp, alpha, stab, C, stride, pad, False)
+    ref = F.conv2d(binsign(x), weight_scale(w) * binsign(w), None,
+                   stride=stride, padding=pad)
+    # integer dot + fp32 scale: exact up to fp32 rounding of alpha*int
+    assert out.shape == ref.shape
+    assert torch.allclose(_cl(out), _cl(ref), atol=1e-3, rtol=1e-4), \
+        (out - ref).abs().max().item()
+
+
+def test_xnor_conv_bf16_out():
+    x = torch.randn(2, 64, 8, 8, device="cuda")
+    w = torch.randn(32, 64, 3, 3, device="cuda")
+    nat = _nat()
+    xp = nat.sign_pack_nhwc(_cl(x))
+    wp, alpha, stab = nat.weight_pack(w)
+    out = nat.xnor_conv_fwd(xp, wp, alpha, stab, 64, 1, 1, True)
+    assert out.dtype == torch.bfloat16
+    ref = F.conv2d(binsign(x), weight_scale(w) * binsign(w), None, 1, 1)
+    assert torch.allclose(out.float(), ref, atol=0.05, rtol=0.02)
+
+
+def test_binary_conv_autograd_gpu_vs_cpu():
+    torch.manual_seed(3)
+    x = (torch.randn(2, 64, 10, 10) * 1.5)
+    w = torch.randn(32, 64, 3, 3) * 1.2
+
+    xg = _cl(x.cuda()).requires_grad_(True)
+    wg = w.cuda().requires_grad_(True)
+    out_g = BinaryConvFunction.apply(xg, wg, 1, 1, "ste", None, None)
+    g = torch.randn_like(out_g)
+    out_g.backward(g)
+
+    xc = x.clone().requires_grad_(True)
+    wc = w.clone().requires_grad_(True)
+    out_c = BinaryConvFunction.apply(xc, wc, 1, 1, "ste", None, None)
+    out_c.backward(g.cpu())
+
+    assert torch.allclose(out_g.cpu(), out_c, atol=1e-3, rtol=1e-4)
+    assert torch.allclose(xg.grad.cpu(), xc.grad, atol=1e-3, rtol=1e-3)
+    assert torch.allclose(wg.grad.cpu(), wc.grad, atol=1e-3, rtol=1e-3)
+
+
+def test_hard_binary_conv_module_gpu():
+    conv = HardBinaryConv(64, 64, 3, 1, 1).cuda()
+    conv = conv.to(memory_format=torch.channels_last)
+    x = _cl(torch.randn(2, 64, 14, 14, device="cuda")).requires_grad_(True)
+    out = conv(x)
+    out.sum().backward()
+    assert conv.weight.grad is not None and x.grad is not None
+    assert torch.isfinite(out).all()
+
+
+# ---------------- kurtosis ----------------
+
+def test_kurtosis_fused_gpu_matches_cpu():
+    torch.manual_seed(4)
+    ws_cpu = [torch.randn(64, 64, 3, 3) for _ in range(3)]
+    tgts = [1.8, 1.4, 1.2]
+    loss_c, kurts_c = kurtosis_loss_fused(
+        [w.clone().requires_grad_(True) for w in ws_cpu], tgts, "sum")
+    ws_gpu = [w.cuda().requires_grad_(True) for w in ws_cpu]
+    loss_g, kurts_g = kurtosis_loss_fused(ws_gpu, tgts, "sum")
+    assert torch.allclose(loss_g.cpu(), loss_c, atol=1e-4, rtol=1e-4)
+    assert torch.allclose(kurts_g.cpu(), kurts_c, atol=1e-4, rtol=1e-4)
+    loss_g.backward()
+    ws_ref = [w.clone().requires_grad_(True) for w in ws_cpu]
+    loss_ref, _ = kurtosis_loss_fused(ws_ref, tgts, "sum")
+    loss_ref.backward()
+    for g_gpu, g_cpu in zip(ws_gpu, ws_ref):
+        assert torch.allclose(g_gpu.grad.cpu(), g_cpu.grad,
+                              atol=1e-5, rtol=1e-4)
+
+
+# ---------------- weight KD ----------------
+
+def test_weight_kd_fused_gpu():
+    from bdbnn_amd.ops.kd import _FusedWeightKD
+    torch.manual_seed(5)
+    ws = [torch.randn(32, 16, 3, 3, device="cuda").requires_grad_(True)
+          for _ in range(2)]
+    wt = [torch.randn(32, 16, 3, 3, device="cuda") for _ in range(2)]
+    loss = _FusedWeightKD.apply(2, *ws, *wt)
+    ref = sum((torch.exp(b) * (b - a)).mean() for a, b in zip(ws, wt))
+    assert torch.allclose(loss, ref, atol=1e-4, rtol=1e-4)
+    loss.backward()
+    for a, b in zip(ws, wt):
+        assert torch.allclose(a.grad, -torch.exp(b) / b.numel(),
+                              atol=1e-6, rtol=1e-5)
+
+
+# ---------------- fused optimizers ----------------
+
+def test_fused_sgd_matches_torch():
+    torch.manual_seed(6)
+    ps = [torch.randn(100, device="cuda").requires_grad_(True),
+          torch.randn(17, 3, 3, 3, device="cuda").requires_grad_(True)]
+    ref = [p.detach().clone().requires_grad_(True) for p in ps]
+    for p, r in zip(ps, ref):
+        g = torch.randn_like(p)
+        p.grad = g.clone()
+        r.grad = g.clone()
+    opt = FusedSGD(ps, lr=0.1, momentum=0.9, weight_decay=1e-4)
+    topt = torch.optim.SGD(ref, lr=0.1, momentum=0.9, weight_decay=1e-4)
+    for _ in range(3):
+        opt.step()
+        topt.step()
+        for p in ps + ref:
+            p.grad = p.grad * 0.9 + 0.01  # evolve grads deterministically
+    for p, r in zip(ps, ref):
+        assert torch.allclose(p, r, atol=1e-5, rtol=1e-5)
+
+
+def test_fused_adam_matches_torch():
+    torch.manual_seed(7)
+    ps = [torch.randn(257, device="cuda").requires_grad_(True)]
+    ref = [p.detach().clone().requires_grad_(True) for p in ps]
+    for p, r in zip(ps, ref):
+        g = torch.randn_like(p)
+        p.grad = g.clone()
+        r.grad = g.clone()
+    opt = FusedAdam(ps, lr=1e-3, weight_decay=1e-4)
+    topt = torch.optim.Adam(ref, lr=1e-3, weight_decay=1e-4)
+    for _ in range(3):
+        opt.step()
+        topt.step()
+    for p, r in zip(ps, ref):
+        assert torch.allclose(p, r, atol=1e-6, rtol=1e-5)
+
+
+# ---------------- end-to-end ----------------
+
+def test_resnet18_gpu_step():
+    from bdbnn_amd.models import imagenet as im
+    m = im.resnet18(False).cuda().to(memory_format=torch.channels_last)
+    x = _cl(torch.randn(4, 3, 64, 64, device="cuda"))
+    y = torch.randint(0, 1000, (4,), device="cuda")
+    out = m(x)
+    loss = torch.nn.functional.cross_entropy(out, y)
+    loss.backward()
+    assert torch.isfinite(loss)
+    grads_ok = [p.grad is not None for p in m.parameters()]
+    assert all(grads_ok)
