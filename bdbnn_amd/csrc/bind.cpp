@@ -28,6 +28,10 @@ void bdbnn_weight_pack(const float*, uint32_t*, float*, float*, int, int,
 void bdbnn_xnor_conv_fwd(const uint32_t*, const uint32_t*, const float*,
                          const float*, void*, bool, int, int, int, int, int,
                          int, int, int, int, int, int, hipStream_t);
+void bdbnn_prelu_fwd(const void*, const float*, void*, int64_t, int, bool,
+                     hipStream_t);
+void bdbnn_prelu_bwd(const void*, const void*, const float*, void*, float*,
+                     int64_t, int, bool, hipStream_t);
 void bdbnn_kurtosis_fwd(const TensorListArg*, const int*, const int64_t*,
                         int, double*, const float*, float*, float*, float*,
                         hipStream_t);
@@ -194,13 +198,44 @@ at::Tensor xnor_conv_fwd(const at::Tensor& xp, const at::Tensor& wp,
   return out;
 }
 
+// ---------------- prelu ----------------
+
+at::Tensor prelu_fwd(const at::Tensor& x, const at::Tensor& a) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4, "prelu_fwd: 4-D CUDA tensor");
+  auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
+  auto af = a.contiguous().to(at::kFloat);
+  int C = (int)x.size(1);
+  TORCH_CHECK(C <= 1024, "prelu: C <= 1024");
+  auto y = at::empty_like(xc, xc.options(), at::MemoryFormat::ChannelsLast);
+  bdbnn_prelu_fwd(xc.data_ptr(), af.data_ptr<float>(), y.data_ptr(),
+                  xc.numel(), C, xc.scalar_type() == at::kBFloat16,
+                  cur_stream());
+  return y;
+}
+
+std::vector<at::Tensor> prelu_bwd(const at::Tensor& g, const at::Tensor& x,
+                                  const at::Tensor& a) {
+  auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
+  auto gc = g.contiguous(at::MemoryFormat::ChannelsLast);
+  TORCH_CHECK(gc.scalar_type() == xc.scalar_type(),
+              "prelu_bwd: grad dtype must match input");
+  auto af = a.contiguous().to(at::kFloat);
+  int C = (int)x.size(1);
+  auto dx = at::empty_like(xc, xc.options(), at::MemoryFormat::ChannelsLast);
+  auto da = at::empty({C}, xc.options().dtype(at::kFloat));
+  bdbnn_prelu_bwd(xc.data_ptr(), gc.data_ptr(), af.data_ptr<float>(),
+                  dx.data_ptr(), da.data_ptr<float>(), xc.numel(), C,
+                  xc.scalar_type() == at::kBFloat16, cur_stream());
+  return {dx, da};
+}
+
 // ---------------- kurtosis ----------------
 
 std::vector<at::Tensor> kurtosis_fwd(const std::vector<at::Tensor>& ws,
                                      const at::Tensor& targets) {
   auto meta = make_meta(ws);
   auto dev = ws[0].device();
-  auto sched = build_schedule(ws, 256 * 1024, dev);
+  auto sched = build_schedule(ws, 32 * 1024, dev);
   int L = (int)ws.size();
   auto opts = ws[0].options();
   auto mom = at::zeros({L, 4}, opts.dtype(at::kDouble));
@@ -222,7 +257,7 @@ std::vector<at::Tensor> kurtosis_bwd(const std::vector<at::Tensor>& ws,
                                      double gscale) {
   auto meta = make_meta(ws);
   auto dev = ws[0].device();
-  auto sched = build_schedule(ws, 256 * 1024, dev);
+  auto sched = build_schedule(ws, 32 * 1024, dev);
   std::vector<at::Tensor> grads;
   for (auto& w : ws)  // preserve_format: grad layout == weight layout
     grads.push_back(at::empty_like(w, w.options(),
@@ -242,7 +277,7 @@ std::vector<at::Tensor> kurtosis_bwd(const std::vector<at::Tensor>& ws,
 at::Tensor weight_kd_fwd(const std::vector<at::Tensor>& ws,
                          const std::vector<at::Tensor>& wt) {
   auto meta = make_meta(ws);
-  auto sched = build_schedule(ws, 256 * 1024, ws[0].device());
+  auto sched = build_schedule(ws, 32 * 1024, ws[0].device());
   auto s_ptr = make_ptrs(ws);
   std::vector<at::Tensor> wt_c;
   for (auto& t : wt) wt_c.push_back(t.contiguous());
@@ -257,7 +292,7 @@ at::Tensor weight_kd_fwd(const std::vector<at::Tensor>& ws,
 std::vector<at::Tensor> weight_kd_bwd(const std::vector<at::Tensor>& wt,
                                       double gscale) {
   auto meta = make_meta(wt);
-  auto sched = build_schedule(wt, 256 * 1024, wt[0].device());
+  auto sched = build_schedule(wt, 32 * 1024, wt[0].device());
   auto t_ptr = make_ptrs(wt);
   std::vector<at::Tensor> grads;
   for (auto& t : wt)
@@ -277,7 +312,7 @@ void fused_sgd(const std::vector<at::Tensor>& p,
                const std::vector<at::Tensor>& buf, double lr, double momentum,
                double wd) {
   auto meta = make_meta(p);
-  auto sched = build_schedule(p, 256 * 512, p[0].device());
+  auto sched = build_schedule(p, 32 * 1024, p[0].device());
   auto pp = make_ptrs(p);
   auto gg = make_ptrs(g);   // python aligns grad layout to the param's
   auto bb = make_ptrs(buf);
@@ -292,7 +327,7 @@ void fused_adam(const std::vector<at::Tensor>& p,
                 const std::vector<at::Tensor>& m2, double lr, double beta1,
                 double beta2, double eps, double wd, double bc1, double bc2) {
   auto meta = make_meta(p);
-  auto sched = build_schedule(p, 256 * 512, p[0].device());
+  auto sched = build_schedule(p, 32 * 1024, p[0].device());
   auto pp = make_ptrs(p);
   auto gg = make_ptrs(g);   // python aligns grad layout to the param's
   auto mm1 = make_ptrs(m1);
@@ -310,6 +345,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("binsign_decode", &binsign_decode, "elementwise +-1 decode");
   m.def("ste_mask_mul", &ste_mask_mul, "quantizer backward mask multiply");
   m.def("xnor_conv_fwd", &xnor_conv_fwd, "bit-packed XNOR+popcount conv");
+  m.def("prelu_fwd", &prelu_fwd, "fused NHWC per-channel PReLU fwd");
+  m.def("prelu_bwd", &prelu_bwd, "fused NHWC per-channel PReLU bwd");
   m.def("kurtosis_fwd", &kurtosis_fwd, "fused multi-tensor kurtosis fwd");
   m.def("kurtosis_bwd", &kurtosis_bwd, "fused multi-tensor kurtosis bwd");
   m.def("weight_kd_fwd", &weight_kd_fwd, "fused weight-space KD fwd");

@@ -9,7 +9,7 @@
 
 struct PtrList2 { float* ptr[BDBNN_MAX_TENSORS]; };
 
-constexpr int64_t KD_CHUNK_ELEMS = 256 * 1024;
+constexpr int64_t KD_CHUNK_ELEMS = 32 * 1024;
 
 __global__ void weight_kd_fwd_kernel(TensorListArg ws_meta, PtrList2 ws,
                                      PtrList2 wt,

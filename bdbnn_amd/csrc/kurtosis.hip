@@ -15,7 +15,7 @@
 // every block covers CHUNK_ELEMS contiguous elements of its tensor.
 #include "common.h"
 
-constexpr int64_t KURT_CHUNK_ELEMS = 256 * 1024;
+constexpr int64_t KURT_CHUNK_ELEMS = 32 * 1024;
 
 struct GradPtrs { float* ptr[BDBNN_MAX_TENSORS]; };
 

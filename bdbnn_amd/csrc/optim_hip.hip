@@ -8,7 +8,7 @@
 
 struct PtrList { float* ptr[BDBNN_MAX_TENSORS]; };
 
-constexpr int64_t OPT_CHUNK_ELEMS = 256 * 512;
+constexpr int64_t OPT_CHUNK_ELEMS = 32 * 1024;
 
 __global__ void fused_sgd_kernel(TensorListArg params_meta, PtrList p,
                                  PtrList g, PtrList buf,

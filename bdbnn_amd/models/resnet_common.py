@@ -26,6 +26,7 @@ from ..ops.binary_conv import (
     HardBinaryConv_cifar,
 )
 from ..ops.binarize import LearnableBias
+from ..ops.activations import ChannelPReLU
 
 
 class RPReLU(nn.Module):
@@ -34,7 +35,7 @@ class RPReLU(nn.Module):
     def __init__(self, channels):
         super().__init__()
         self.move1 = LearnableBias(channels)
-        self.prelu = nn.PReLU(channels)
+        self.prelu = ChannelPReLU(channels)
         self.move2 = LearnableBias(channels)
 
     def forward(self, x):
@@ -45,7 +46,7 @@ def _make_act(kind, channels):
     if kind == "relu":
         return nn.ReLU(inplace=True)
     if kind == "prelu":
-        return nn.PReLU(channels)
+        return ChannelPReLU(channels)
     if kind == "rprelu":
         return RPReLU(channels)
     raise ValueError(kind)

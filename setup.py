@@ -21,6 +21,7 @@ CSRC = os.path.join(ROOT, "bdbnn_amd", "csrc")
 sources = [os.path.join(CSRC, f) for f in (
     "bind.cpp",
     "pack.hip",
+    "prelu.hip",
     "xnor_conv.hip",
     "kurtosis.hip",
     "kd.hip",

@@ -78,3 +78,22 @@ def test_learnable_bias_shape():
     x = torch.randn(2, 8, 4, 4)
     assert lb(x).shape == x.shape
     assert lb.bias.shape == (1, 8, 1, 1)
+
+
+def test_channel_prelu_cpu_matches_torch():
+    from bdbnn_amd.ops.activations import ChannelPReLU
+    torch.manual_seed(9)
+    m = ChannelPReLU(8)
+    with torch.no_grad():
+        m.weight.uniform_(-0.5, 0.5)
+    ref = torch.nn.PReLU(8)
+    with torch.no_grad():
+        ref.weight.copy_(m.weight)
+    x = torch.randn(2, 8, 5, 5, requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    out = m(x); out2 = ref(x2)
+    assert torch.allclose(out, out2)
+    g = torch.randn_like(out)
+    out.backward(g); out2.backward(g)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-6)
+    assert torch.allclose(m.weight.grad, ref.weight.grad, atol=1e-5)

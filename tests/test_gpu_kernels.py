@@ -232,3 +232,33 @@ def test_resnet18_gpu_step():
     assert torch.isfinite(loss)
     grads_ok = [p.grad is not None for p in m.parameters()]
     assert all(grads_ok)
+
+
+# ---------------- fused PReLU ----------------
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_channel_prelu_fused(dtype):
+    from bdbnn_amd.ops.activations import ChannelPReLU
+    torch.manual_seed(8)
+    m = ChannelPReLU(32).cuda()
+    with torch.no_grad():
+        m.weight.uniform_(-0.5, 0.5)
+    x = (torch.randn(4, 32, 9, 9, device="cuda", dtype=dtype) * 2)
+    x = _cl(x).requires_grad_(True)
+    out = m(x)
+    g = torch.randn_like(out)
+    out.backward(g)
+
+    ref_m = torch.nn.PReLU(32).cuda()
+    with torch.no_grad():
+        ref_m.weight.copy_(m.weight)
+    x2 = x.detach().clone().requires_grad_(True)
+    out2 = ref_m(x2.float())
+    out2.backward(g.float())
+
+    atol = 1e-5 if dtype == torch.float32 else 0.05
+    assert torch.allclose(out.float(), out2, atol=atol, rtol=1e-2)
+    assert torch.allclose(x.grad.float(), x2.grad, atol=atol, rtol=1e-2)
+    assert torch.allclose(m.weight.grad, ref_m.weight.grad,
+                          atol=0.05 if dtype == torch.bfloat16 else 1e-3,
+                          rtol=1e-2)
