@@ -32,6 +32,23 @@ void bdbnn_prelu_fwd(const void*, const float*, void*, int64_t, int, bool,
                      hipStream_t);
 void bdbnn_prelu_bwd(const void*, const void*, const float*, void*, float*,
                      int64_t, int, bool, hipStream_t);
+void bdbnn_bn_stats(const void*, float*, float*, int64_t, int, bool,
+                    hipStream_t);
+void bdbnn_bn_finalize(const float*, const float*, float*, float*, float*,
+                       float*, int, float, float, float, hipStream_t);
+void bdbnn_bn_act_fwd(const void*, const void*, const float*, const float*,
+                      const float*, const float*, const float*, void*, void*,
+                      int64_t, int, int, bool, hipStream_t);
+void bdbnn_bn_act_bwd_reduce(const void*, const void*, const void*,
+                             const float*, const float*, const float*,
+                             float*, int64_t, int, int, bool, hipStream_t);
+void bdbnn_bn_act_bwd_apply(const void*, const void*, const void*,
+                            const float*, const float*, const float*,
+                            const float*, const float*, void*, void*,
+                            int64_t, int, int, float, bool, hipStream_t);
+void bdbnn_bn_act_eval(const void*, const void*, const float*, const float*,
+                       const float*, const float*, const float*, void*,
+                       int64_t, int, int, float, bool, hipStream_t);
 void bdbnn_kurtosis_fwd(const TensorListArg*, const int*, const int64_t*,
                         int, double*, const float*, float*, float*, float*,
                         hipStream_t);
@@ -229,6 +246,132 @@ std::vector<at::Tensor> prelu_bwd(const at::Tensor& g, const at::Tensor& x,
   return {dx, da};
 }
 
+// ---------------- fused BN (+add) (+act) ----------------
+
+std::vector<at::Tensor> bn_act_fwd_train(
+    const at::Tensor& x, const c10::optional<at::Tensor>& skip,
+    const at::Tensor& gamma, const at::Tensor& beta,
+    const c10::optional<at::Tensor>& a,
+    c10::optional<at::Tensor> running_mean,
+    c10::optional<at::Tensor> running_var, double momentum, double eps,
+    int64_t act_kind) {
+  auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
+  int C = (int)x.size(1);
+  TORCH_CHECK(C <= 1024, "fused bn: C <= 1024");
+  int64_t n = xc.numel();
+  bool bf16 = is_bf16(xc);
+  auto fopt = xc.options().dtype(at::kFloat);
+  auto s1 = at::empty({C}, fopt);
+  auto s2 = at::empty({C}, fopt);
+  auto mean = at::empty({C}, fopt);
+  auto invstd = at::empty({C}, fopt);
+  bdbnn_bn_stats(xc.data_ptr(), s1.data_ptr<float>(), s2.data_ptr<float>(),
+                 n, C, bf16, cur_stream());
+  float* rm = running_mean.has_value()
+                  ? running_mean->data_ptr<float>() : nullptr;
+  float* rv = running_var.has_value()
+                  ? running_var->data_ptr<float>() : nullptr;
+  bdbnn_bn_finalize(s1.data_ptr<float>(), s2.data_ptr<float>(),
+                    mean.data_ptr<float>(), invstd.data_ptr<float>(), rm, rv,
+                    C, (float)(n / C), (float)momentum, (float)eps,
+                    cur_stream());
+  auto out = at::empty_like(xc, xc.options(), at::MemoryFormat::ChannelsLast);
+  auto z = at::empty_like(xc, xc.options(), at::MemoryFormat::ChannelsLast);
+  at::Tensor skipc;
+  const void* skip_ptr = nullptr;
+  if (skip.has_value()) {
+    skipc = skip->to(xc.scalar_type())
+                .contiguous(at::MemoryFormat::ChannelsLast);
+    skip_ptr = skipc.data_ptr();
+  }
+  auto gf = gamma.contiguous().to(at::kFloat);
+  auto bf = beta.contiguous().to(at::kFloat);
+  at::Tensor af;
+  const float* a_ptr = nullptr;
+  if (a.has_value()) { af = a->contiguous().to(at::kFloat);
+                       a_ptr = af.data_ptr<float>(); }
+  bdbnn_bn_act_fwd(xc.data_ptr(), skip_ptr, mean.data_ptr<float>(),
+                   invstd.data_ptr<float>(), gf.data_ptr<float>(),
+                   bf.data_ptr<float>(), a_ptr, out.data_ptr(), z.data_ptr(),
+                   n, C, (int)act_kind, bf16, cur_stream());
+  return {out, z, mean, invstd};
+}
+
+std::vector<at::Tensor> bn_act_bwd(
+    const at::Tensor& dy, const at::Tensor& z, const at::Tensor& x,
+    const at::Tensor& mean, const at::Tensor& invstd,
+    const at::Tensor& gamma, const c10::optional<at::Tensor>& a,
+    int64_t act_kind, bool need_dskip) {
+  auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
+  auto zc = z.contiguous(at::MemoryFormat::ChannelsLast);
+  auto dyc = dy.to(xc.scalar_type()).contiguous(at::MemoryFormat::ChannelsLast);
+  int C = (int)x.size(1);
+  int64_t n = xc.numel();
+  bool bf16 = is_bf16(xc);
+  auto fopt = xc.options().dtype(at::kFloat);
+  auto sums = at::empty({C, 3}, fopt);
+  at::Tensor af;
+  const float* a_ptr = nullptr;
+  if (a.has_value()) { af = a->contiguous().to(at::kFloat);
+                       a_ptr = af.data_ptr<float>(); }
+  bdbnn_bn_act_bwd_reduce(dyc.data_ptr(), zc.data_ptr(), xc.data_ptr(),
+                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                          a_ptr, sums.data_ptr<float>(), n, C, (int)act_kind,
+                          bf16, cur_stream());
+  auto gf = gamma.contiguous().to(at::kFloat);
+  auto dx = at::empty_like(xc, xc.options(), at::MemoryFormat::ChannelsLast);
+  at::Tensor dskip;
+  void* dskip_ptr = nullptr;
+  if (need_dskip) {
+    dskip = at::empty_like(xc, xc.options(), at::MemoryFormat::ChannelsLast);
+    dskip_ptr = dskip.data_ptr();
+  } else {
+    dskip = at::empty({0}, xc.options());
+  }
+  bdbnn_bn_act_bwd_apply(dyc.data_ptr(), zc.data_ptr(), xc.data_ptr(),
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         gf.data_ptr<float>(), a_ptr, sums.data_ptr<float>(),
+                         dx.data_ptr(), dskip_ptr, n, C, (int)act_kind,
+                         (float)(1.0 / (double)(n / C)), bf16, cur_stream());
+  auto dbeta = sums.select(1, 0).clone();
+  auto dgamma = sums.select(1, 1).clone();
+  auto da = sums.select(1, 2).clone();
+  return {dx, dskip, dgamma, dbeta, da};
+}
+
+at::Tensor bn_act_eval(const at::Tensor& x,
+                       const c10::optional<at::Tensor>& skip,
+                       const at::Tensor& gamma, const at::Tensor& beta,
+                       const c10::optional<at::Tensor>& a,
+                       const at::Tensor& rm, const at::Tensor& rv,
+                       double eps, int64_t act_kind) {
+  auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
+  int C = (int)x.size(1);
+  int64_t n = xc.numel();
+  bool bf16 = is_bf16(xc);
+  auto out = at::empty_like(xc, xc.options(), at::MemoryFormat::ChannelsLast);
+  at::Tensor skipc;
+  const void* skip_ptr = nullptr;
+  if (skip.has_value()) {
+    skipc = skip->to(xc.scalar_type())
+                .contiguous(at::MemoryFormat::ChannelsLast);
+    skip_ptr = skipc.data_ptr();
+  }
+  auto gf = gamma.contiguous().to(at::kFloat);
+  auto bf = beta.contiguous().to(at::kFloat);
+  auto rmf = rm.contiguous().to(at::kFloat);
+  auto rvf = rv.contiguous().to(at::kFloat);
+  at::Tensor af;
+  const float* a_ptr = nullptr;
+  if (a.has_value()) { af = a->contiguous().to(at::kFloat);
+                       a_ptr = af.data_ptr<float>(); }
+  bdbnn_bn_act_eval(xc.data_ptr(), skip_ptr, rmf.data_ptr<float>(),
+                    rvf.data_ptr<float>(), gf.data_ptr<float>(),
+                    bf.data_ptr<float>(), a_ptr, out.data_ptr(), n, C,
+                    (int)act_kind, (float)eps, bf16, cur_stream());
+  return out;
+}
+
 // ---------------- kurtosis ----------------
 
 std::vector<at::Tensor> kurtosis_fwd(const std::vector<at::Tensor>& ws,
@@ -347,6 +490,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("xnor_conv_fwd", &xnor_conv_fwd, "bit-packed XNOR+popcount conv");
   m.def("prelu_fwd", &prelu_fwd, "fused NHWC per-channel PReLU fwd");
   m.def("prelu_bwd", &prelu_bwd, "fused NHWC per-channel PReLU bwd");
+  m.def("bn_act_fwd_train", &bn_act_fwd_train,
+        "fused BN(+add)(+act) training forward");
+  m.def("bn_act_bwd", &bn_act_bwd, "fused BN(+add)(+act) backward");
+  m.def("bn_act_eval", &bn_act_eval, "fused BN(+add)(+act) eval forward");
   m.def("kurtosis_fwd", &kurtosis_fwd, "fused multi-tensor kurtosis fwd");
   m.def("kurtosis_bwd", &kurtosis_bwd, "fused multi-tensor kurtosis bwd");
   m.def("weight_kd_fwd", &weight_kd_fwd, "fused weight-space KD fwd");

@@ -27,6 +27,14 @@ from ..ops.binary_conv import (
 )
 from ..ops.binarize import LearnableBias
 from ..ops.activations import ChannelPReLU
+from ..ops.bn_act import fused_bn_act
+
+
+class FusedDownsample(nn.Sequential):
+    """conv + BN downsample with the fused BN kernel (names '0'/'1' kept)."""
+
+    def forward(self, x):
+        return fused_bn_act(self[0](x), self[1])
 
 
 class RPReLU(nn.Module):
@@ -69,9 +77,8 @@ class BasicBlock(nn.Module):
 
     def forward(self, x):
         identity = self.downsample(x) if self.downsample is not None else x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.bn2(self.conv2(out))
-        out = self.relu(out + identity)
+        out = fused_bn_act(self.conv1(x), self.bn1, "relu")
+        out = fused_bn_act(self.conv2(out), self.bn2, "relu", skip=identity)
         return out
 
 
@@ -98,10 +105,8 @@ class BiBasicBlock(nn.Module):
 
     def forward(self, x):
         identity = self.downsample(x) if self.downsample is not None else x
-        out = self.bn1(self.conv1(x))
-        out = self.act1(out + identity)
-        out2 = self.bn2(self.conv2(out))
-        out = self.act2(out2 + out)
+        out = fused_bn_act(self.conv1(x), self.bn1, self.act1, skip=identity)
+        out = fused_bn_act(self.conv2(out), self.bn2, self.act2, skip=out)
         return out
 
 
@@ -144,7 +149,7 @@ class ResNet(nn.Module):
                 ds_conv = conv_cls(self.inplanes, planes, 1, stride, 0)
             else:
                 ds_conv = nn.Conv2d(self.inplanes, planes, 1, stride, bias=False)
-            downsample = nn.Sequential(ds_conv, nn.BatchNorm2d(planes))
+            downsample = FusedDownsample(ds_conv, nn.BatchNorm2d(planes))
         layers = []
         kw = dict(conv_cls=conv_cls, act=act) if self.binary else {}
         layers.append(block_fn(self.inplanes, planes, stride, downsample, **kw))
@@ -154,7 +159,7 @@ class ResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.maxpool(fused_bn_act(self.conv1(x), self.bn1, "relu"))
         x = self.layer1(x)
         x = self.layer2(x)
         x = self.layer3(x)
@@ -222,7 +227,7 @@ class CifarResNet(nn.Module):
                 ds_conv = kw["conv_cls"](self.inplanes, planes, 1, stride, 0)
             else:
                 ds_conv = nn.Conv2d(self.inplanes, planes, 1, stride, bias=False)
-            downsample = nn.Sequential(ds_conv, nn.BatchNorm2d(planes))
+            downsample = FusedDownsample(ds_conv, nn.BatchNorm2d(planes))
         layers = [block_fn(self.inplanes, planes, stride, downsample, **kw)]
         self.inplanes = planes
         for _ in range(1, blocks):
@@ -230,7 +235,7 @@ class CifarResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        x = self.relu(self.bn1(self.conv1(x)))
+        x = fused_bn_act(self.conv1(x), self.bn1, "relu")
         x = self.layer1(x)
         x = self.layer2(x)
         x = self.layer3(x)

@@ -1,0 +1,89 @@
+"""Fused BatchNorm (+ residual add) (+ activation) — K8.
+
+``fused_bn_act(x, bn, act=None, skip=None)`` runs the whole
+conv-output -> BN -> (+skip) -> PReLU/ReLU tail of a BD-BNN block as two
+kernels forward (stats + normalize/add/act) and two backward (reduce +
+apply) on the GPU, replacing the MIOpen BN pipeline + separate add +
+activation (profiles/r01_bench_b256_kernel_stats.md).  Parameters stay
+in the ordinary ``nn.BatchNorm2d`` / ``ChannelPReLU`` modules, so
+checkpoints are unchanged.
+
+CPU (and any non-channels_last corner): exact composition fallback.
+"""
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .. import _C
+from .activations import ChannelPReLU
+
+_ACT_NONE, _ACT_PRELU, _ACT_RELU = 0, 1, 2
+
+
+class _BNActFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, skip, gamma, beta, a, running_mean, running_var,
+                momentum, eps, act_kind, training):
+        nat = _C.native_required()
+        out, z, mean, invstd = nat.bn_act_fwd_train(
+            x, skip, gamma, beta, a, running_mean, running_var,
+            momentum, eps, act_kind)
+        ctx.save_for_backward(x, z, mean, invstd, gamma,
+                              a if a is not None else torch.empty(0))
+        ctx.act_kind = act_kind
+        ctx.has_skip = skip is not None
+        ctx.has_a = a is not None
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, z, mean, invstd, gamma, a = ctx.saved_tensors
+        nat = _C.native_required()
+        dx, dskip, dgamma, dbeta, da = nat.bn_act_bwd(
+            dy, z, x, mean, invstd, gamma,
+            a if ctx.has_a else None, ctx.act_kind, ctx.has_skip)
+        return (dx,
+                dskip if ctx.has_skip else None,
+                dgamma, dbeta,
+                da if ctx.has_a else None,
+                None, None, None, None, None, None)
+
+
+def fused_bn_act(x, bn: nn.BatchNorm2d, act=None, skip=None):
+    """BN(x) (+skip) then act.  act: None | ChannelPReLU | 'relu'."""
+    if isinstance(act, ChannelPReLU):
+        act_kind, a = _ACT_PRELU, act.weight
+    elif act == "relu":
+        act_kind, a = _ACT_RELU, None
+    elif act is None:
+        act_kind, a = _ACT_NONE, None
+    else:  # generic module: apply unfused after BN+add
+        out = fused_bn_act(x, bn, None, skip)
+        return act(out)
+
+    use_fused = (x.is_cuda and x.dim() == 4 and x.size(1) <= 1024
+                 and _C.has_native())
+    if use_fused and bn.training:
+        if bn.track_running_stats and bn.num_batches_tracked is not None:
+            bn.num_batches_tracked.add_(1)
+        return _BNActFn.apply(
+            x, skip, bn.weight, bn.bias, a, bn.running_mean, bn.running_var,
+            bn.momentum if bn.momentum is not None else 0.1, bn.eps,
+            act_kind, True)
+    if use_fused and not torch.is_grad_enabled() \
+            and bn.running_mean is not None:
+        nat = _C.native_required()
+        return nat.bn_act_eval(x, skip, bn.weight, bn.bias, a,
+                               bn.running_mean, bn.running_var, bn.eps,
+                               act_kind)
+
+    # composition fallback (CPU / oracle)
+    z = bn(x)
+    if skip is not None:
+        z = z + skip
+    if act_kind == _ACT_PRELU:
+        return act(z)
+    if act_kind == _ACT_RELU:
+        return F.relu(z)
+    return z

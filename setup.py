@@ -22,6 +22,7 @@ sources = [os.path.join(CSRC, f) for f in (
     "bind.cpp",
     "pack.hip",
     "prelu.hip",
+    "bn_act.hip",
     "xnor_conv.hip",
     "kurtosis.hip",
     "kd.hip",

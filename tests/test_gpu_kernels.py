@@ -258,7 +258,87 @@ def test_channel_prelu_fused(dtype):
 
     atol = 1e-5 if dtype == torch.float32 else 0.05
     assert torch.allclose(out.float(), out2, atol=atol, rtol=1e-2)
-    assert torch.allclose(x.grad.float(), x2.grad, atol=atol, rtol=1e-2)
+    assert torch.allclose(x.grad.float(), x2.grad.float(), atol=atol,
+                          rtol=1e-2)
     assert torch.allclose(m.weight.grad, ref_m.weight.grad,
                           atol=0.05 if dtype == torch.bfloat16 else 1e-3,
                           rtol=1e-2)
+
+
+# ---------------- fused BN (+add) (+act) ----------------
+
+@pytest.mark.parametrize("act_kind", ["prelu", "relu", "none"])
+@pytest.mark.parametrize("with_skip", [False, True])
+def test_fused_bn_act_train(act_kind, with_skip):
+    from bdbnn_amd.ops.bn_act import fused_bn_act
+    from bdbnn_amd.ops.activations import ChannelPReLU
+    torch.manual_seed(10)
+    C = 32
+    bn = torch.nn.BatchNorm2d(C).cuda()
+    bn2 = torch.nn.BatchNorm2d(C).cuda()
+    with torch.no_grad():
+        bn.weight.uniform_(0.5, 1.5); bn.bias.uniform_(-0.3, 0.3)
+        bn2.load_state_dict(bn.state_dict())
+    act = act2 = None
+    if act_kind == "prelu":
+        act = ChannelPReLU(C).cuda()
+        act2 = torch.nn.PReLU(C).cuda()
+        with torch.no_grad():
+            act.weight.uniform_(-0.4, 0.6)
+            act2.weight.copy_(act.weight)
+    elif act_kind == "relu":
+        act = act2 = "relu"
+    x = _cl(torch.randn(8, C, 14, 14, device="cuda") * 2).requires_grad_(True)
+    skip = (_cl(torch.randn(8, C, 14, 14, device="cuda")).requires_grad_(True)
+            if with_skip else None)
+    bn.train(); bn2.train()
+    out = fused_bn_act(x, bn, act, skip=skip)
+
+    # reference composition
+    x2 = x.detach().clone().requires_grad_(True)
+    skip2 = skip.detach().clone().requires_grad_(True) if with_skip else None
+    z = bn2(x2)
+    if with_skip:
+        z = z + skip2
+    if act_kind == "prelu":
+        ref = act2(z)
+    elif act_kind == "relu":
+        ref = torch.relu(z)
+    else:
+        ref = z
+    assert torch.allclose(out, _cl(ref), atol=2e-4, rtol=1e-4), \
+        (out - _cl(ref)).abs().max().item()
+
+    g = torch.randn_like(out)
+    out.backward(g)
+    ref.backward(_cl(g))
+    assert torch.allclose(x.grad, x2.grad, atol=2e-4, rtol=1e-3)
+    if with_skip:
+        assert torch.allclose(skip.grad, skip2.grad, atol=2e-4, rtol=1e-3)
+    assert torch.allclose(bn.weight.grad, bn2.weight.grad, atol=2e-3,
+                          rtol=1e-3)
+    assert torch.allclose(bn.bias.grad, bn2.bias.grad, atol=2e-3, rtol=1e-3)
+    if act_kind == "prelu":
+        assert torch.allclose(act.weight.grad, act2.weight.grad, atol=2e-3,
+                              rtol=1e-3)
+    # running stats updated identically
+    assert torch.allclose(bn.running_mean, bn2.running_mean, atol=1e-4)
+    assert torch.allclose(bn.running_var, bn2.running_var, atol=1e-4)
+    assert bn.num_batches_tracked.item() == bn2.num_batches_tracked.item()
+
+
+def test_fused_bn_act_eval():
+    from bdbnn_amd.ops.bn_act import fused_bn_act
+    torch.manual_seed(11)
+    C = 16
+    bn = torch.nn.BatchNorm2d(C).cuda()
+    with torch.no_grad():
+        bn.running_mean.uniform_(-1, 1)
+        bn.running_var.uniform_(0.5, 2)
+        bn.weight.uniform_(0.5, 1.5)
+    bn.eval()
+    x = _cl(torch.randn(4, C, 7, 7, device="cuda"))
+    with torch.no_grad():
+        out = fused_bn_act(x, bn, "relu")
+        ref = torch.relu(bn(x))
+    assert torch.allclose(out, _cl(ref), atol=1e-4, rtol=1e-4)
