@@ -15,16 +15,22 @@ from .. import _C
 
 class _ChannelPReLUFn(torch.autograd.Function):
     @staticmethod
+    def _fused_ok(x):
+        C = x.size(1)
+        return (x.is_cuda and x.dim() == 4 and C <= 1024
+                and (C & (C - 1)) == 0)
+
+    @staticmethod
     def forward(ctx, x, w):
         ctx.save_for_backward(x, w)
-        if x.is_cuda and x.dim() == 4 and x.size(1) <= 1024:
+        if _ChannelPReLUFn._fused_ok(x):
             return _C.native_required().prelu_fwd(x, w)
         return F.prelu(x, w.to(x.dtype))
 
     @staticmethod
     def backward(ctx, g):
         x, w = ctx.saved_tensors
-        if x.is_cuda and x.dim() == 4 and x.size(1) <= 1024:
+        if _ChannelPReLUFn._fused_ok(x):
             dx, da = _C.native_required().prelu_bwd(g.to(x.dtype), x, w)
             return dx, da.to(w.dtype)
         mask = x > 0

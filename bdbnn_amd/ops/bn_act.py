@@ -62,8 +62,9 @@ def fused_bn_act(x, bn: nn.BatchNorm2d, act=None, skip=None):
         out = fused_bn_act(x, bn, None, skip)
         return act(out)
 
-    use_fused = (x.is_cuda and x.dim() == 4 and x.size(1) <= 1024
-                 and _C.has_native())
+    C = x.size(1) if x.dim() == 4 else 0
+    use_fused = (x.is_cuda and x.dim() == 4 and 0 < C <= 1024
+                 and (C & (C - 1)) == 0 and _C.has_native())
     if use_fused and bn.training:
         if bn.track_running_stats and bn.num_batches_tracked is not None:
             bn.num_batches_tracked.add_(1)
