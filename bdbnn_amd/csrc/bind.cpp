@@ -165,6 +165,7 @@ std::vector<at::Tensor> weight_pack(const at::Tensor& w) {
 
 at::Tensor binsign_decode(const at::Tensor& x, bool out_bf16) {
   TORCH_CHECK(x.is_cuda(), "binsign_decode: CUDA tensor");
+  TORCH_CHECK(x.numel() % 8 == 0, "binsign_decode: numel % 8 == 0");
   auto fmt = x.dim() == 4 ? at::MemoryFormat::ChannelsLast
                           : at::MemoryFormat::Contiguous;
   auto xc = x.contiguous(fmt);
@@ -179,6 +180,7 @@ at::Tensor ste_mask_mul(const at::Tensor& g, const at::Tensor& x, int mode,
                         double t, double k) {
   TORCH_CHECK(g.is_cuda() && x.is_cuda() && g.numel() == x.numel(),
               "ste_mask_mul: matching CUDA tensors");
+  TORCH_CHECK(x.numel() % 8 == 0, "ste_mask_mul: numel % 8 == 0");
   auto fmt = x.dim() == 4 ? at::MemoryFormat::ChannelsLast
                           : at::MemoryFormat::Contiguous;
   auto xc = x.contiguous(fmt);

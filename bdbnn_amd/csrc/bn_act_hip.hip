@@ -7,29 +7,19 @@
 //   bwd: reduce pass (reads dy,z,x -> per-channel sums + da) +
 //        apply pass (reads dy,z,x -> writes dx and dskip)
 //
-// Channel-reduction layout: C is a power of two <= 1024 (python guards);
-// every thread owns FIXED channels (c = tid mod C), accumulates in
-// registers over its pixel stripe — coalesced along C, no per-element
-// atomics — and lands ONE LDS add + ONE global atomic per owned channel.
+// Memory layout: every thread owns 8 consecutive channels (16-B vector
+// loads, guide G13), accumulates channel sums in registers over its pixel
+// stripe, and lands one LDS add + one global atomic per owned channel.
+// C must be a power of two, 8 <= C <= 1024 (python guards; fallback is
+// the composition path).
 //
 // BN semantics match nn.BatchNorm2d: biased batch var for normalization,
 // unbiased var into running_var, momentum update in the finalize step.
 // act_kind: 0 = identity, 1 = per-channel PReLU, 2 = ReLU.
 #include "common.h"
+#include "vec8.h"
 
-#define BN_MAXC_PER_THREAD 4  // supports C up to 4*256 = 1024
-
-__device__ __forceinline__ float load_f(const void* p, int64_t i, bool bf16) {
-  return bf16 ? bf16_to_f32(((const uint16_t*)p)[i]) : ((const float*)p)[i];
-}
-
-__device__ __forceinline__ void store_f(void* p, int64_t i, float v,
-                                        bool bf16) {
-  if (bf16) ((uint16_t*)p)[i] = f32_to_bf16(v);
-  else      ((float*)p)[i] = v;
-}
-
-// ---- pass 1: per-channel sum / sumsq (fixed-channel ownership) ----
+// ---- pass 1: per-channel sum / sumsq ----
 template <typename T>
 __global__ void bn_stats_kernel(const T* __restrict__ x,
                                 float* __restrict__ s1,
@@ -40,31 +30,18 @@ __global__ void bn_stats_kernel(const T* __restrict__ x,
   float* l2 = l1 + C;
   for (int c = threadIdx.x; c < C; c += blockDim.x) { l1[c] = 0.f; l2[c] = 0.f; }
   __syncthreads();
-  const bool bf16 = sizeof(T) == 2;
-  if (C <= 256) {
-    int rows = 256 / C;
-    int c = threadIdx.x & (C - 1);
-    int r_off = threadIdx.x / C;
-    float a1 = 0.f, a2 = 0.f;
-    for (int64_t p = (int64_t)blockIdx.x * rows + r_off; p < n_pix;
-         p += (int64_t)gridDim.x * rows) {
-      float v = load_f(x, p * C + c, bf16);
-      a1 += v; a2 += v * v;
-    }
-    atomicAdd(&l1[c], a1);
-    atomicAdd(&l2[c], a2);
-  } else {
-    int per = C / 256;
-    float a1[BN_MAXC_PER_THREAD] = {}, a2[BN_MAXC_PER_THREAD] = {};
-    for (int64_t p = blockIdx.x; p < n_pix; p += gridDim.x)
-      for (int j = 0; j < per; ++j) {
-        float v = load_f(x, p * C + threadIdx.x + j * 256, bf16);
-        a1[j] += v; a2[j] += v * v;
-      }
-    for (int j = 0; j < per; ++j) {
-      l1[threadIdx.x + j * 256] = a1[j];
-      l2[threadIdx.x + j * 256] = a2[j];
-    }
+  ChanMap m = chan_map8(C);
+  float a1[8] = {}, a2[8] = {};
+  float v[8];
+  for (int64_t p = m.p0; p < n_pix; p += m.pstep) {
+    load8(x, p * C + m.c0, v);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { a1[j] += v[j]; a2[j] += v[j] * v[j]; }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    atomicAdd(&l1[m.c0 + j], a1[j]);
+    atomicAdd(&l2[m.c0 + j], a2[j]);
   }
   __syncthreads();
   for (int c = threadIdx.x; c < C; c += blockDim.x) {
@@ -94,7 +71,7 @@ __global__ void bn_finalize_kernel(const float* __restrict__ s1,
   }
 }
 
-// ---- pass 2: normalize + add + act (fixed-channel, scale/shift hoisted) ----
+// ---- pass 2: normalize + add + act ----
 template <typename T, int ACT>
 __global__ void bn_act_fwd_kernel(const T* __restrict__ x,
                                   const T* __restrict__ skip,
@@ -105,27 +82,30 @@ __global__ void bn_act_fwd_kernel(const T* __restrict__ x,
                                   const float* __restrict__ a,
                                   T* __restrict__ out, T* __restrict__ zout,
                                   int64_t n_pix, int C) {
-  const bool bf16 = sizeof(T) == 2;
-  int c, r_off, rows, per;
-  if (C <= 256) { rows = 256 / C; c = threadIdx.x & (C - 1);
-                  r_off = threadIdx.x / C; per = 1; }
-  else { rows = 1; c = threadIdx.x; r_off = 0; per = C / 256; }
-  for (int j = 0; j < per; ++j) {
-    int cc = c + j * 256;
-    float sc = gamma[cc] * invstd[cc];
-    float sh = beta[cc] - mean[cc] * sc;
-    float av = (ACT == 1) ? a[cc] : 0.f;
-    for (int64_t p = (int64_t)blockIdx.x * rows + r_off; p < n_pix;
-         p += (int64_t)gridDim.x * rows) {
-      int64_t i = p * C + cc;
-      float z = sc * load_f(x, i, bf16) + sh;
-      if (skip != nullptr) z += load_f(skip, i, bf16);
-      float o = z;
-      if (ACT == 1) o = z > 0.f ? z : av * z;
-      else if (ACT == 2) o = fmaxf(z, 0.f);
-      store_f(out, i, o, bf16);
-      if (zout != nullptr) store_f(zout, i, z, bf16);
+  ChanMap m = chan_map8(C);
+  float sc[8], sh[8], av[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int cc = m.c0 + j;
+    sc[j] = gamma[cc] * invstd[cc];
+    sh[j] = beta[cc] - mean[cc] * sc[j];
+    av[j] = (ACT == 1) ? a[cc] : 0.f;
+  }
+  float v[8], s[8], z[8], o[8];
+  for (int64_t p = m.p0; p < n_pix; p += m.pstep) {
+    int64_t i = p * C + m.c0;
+    load8(x, i, v);
+    if (skip != nullptr) load8(skip, i, s);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      z[j] = sc[j] * v[j] + sh[j];
+      if (skip != nullptr) z[j] += s[j];
+      o[j] = z[j];
+      if (ACT == 1) o[j] = z[j] > 0.f ? z[j] : av[j] * z[j];
+      else if (ACT == 2) o[j] = fmaxf(z[j], 0.f);
     }
+    store8(out, i, o);
+    if (zout != nullptr) store8(zout, i, z);
   }
 }
 
@@ -145,36 +125,40 @@ __global__ void bn_act_bwd_reduce_kernel(
     r0[c] = 0.f; r1[c] = 0.f; r2[c] = 0.f;
   }
   __syncthreads();
-  const bool bf16 = sizeof(T) == 2;
-  int c, r_off, rows, per;
-  if (C <= 256) { rows = 256 / C; c = threadIdx.x & (C - 1);
-                  r_off = threadIdx.x / C; per = 1; }
-  else { rows = 1; c = threadIdx.x; r_off = 0; per = C / 256; }
-  for (int j = 0; j < per; ++j) {
-    int cc = c + j * 256;
-    float mu = mean[cc], is = invstd[cc];
-    float av = (ACT == 1) ? a[cc] : 0.f;
-    float s0 = 0.f, s1 = 0.f, s2 = 0.f;
-    for (int64_t p = (int64_t)blockIdx.x * rows + r_off; p < n_pix;
-         p += (int64_t)gridDim.x * rows) {
-      int64_t i = p * C + cc;
-      float dyi = load_f(dy, i, bf16);
-      float dz = dyi;
+  ChanMap m = chan_map8(C);
+  float mu[8], is[8], av[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int cc = m.c0 + j;
+    mu[j] = mean[cc]; is[j] = invstd[cc];
+    av[j] = (ACT == 1) ? a[cc] : 0.f;
+  }
+  float s0[8] = {}, s1[8] = {}, s2[8] = {};
+  float dyv[8], zv[8], xv[8];
+  for (int64_t p = m.p0; p < n_pix; p += m.pstep) {
+    int64_t i = p * C + m.c0;
+    load8(dy, i, dyv);
+    load8(x, i, xv);
+    if (ACT != 0) load8(z, i, zv);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float dz = dyv[j];
       if (ACT == 1) {
-        float zi = load_f(z, i, bf16);
-        dz = zi > 0.f ? dyi : av * dyi;
-        if (zi <= 0.f) s2 += dyi * zi;
+        dz = zv[j] > 0.f ? dyv[j] : av[j] * dyv[j];
+        if (zv[j] <= 0.f) s2[j] += dyv[j] * zv[j];
       } else if (ACT == 2) {
-        float zi = load_f(z, i, bf16);
-        dz = zi > 0.f ? dyi : 0.f;
+        dz = zv[j] > 0.f ? dyv[j] : 0.f;
       }
-      float xhat = (load_f(x, i, bf16) - mu) * is;
-      s0 += dz;
-      s1 += dz * xhat;
+      float xhat = (xv[j] - mu[j]) * is[j];
+      s0[j] += dz;
+      s1[j] += dz * xhat;
     }
-    atomicAdd(&r0[cc], s0);
-    atomicAdd(&r1[cc], s1);
-    if (ACT == 1) atomicAdd(&r2[cc], s2);
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    atomicAdd(&r0[m.c0 + j], s0[j]);
+    atomicAdd(&r1[m.c0 + j], s1[j]);
+    if (ACT == 1) atomicAdd(&r2[m.c0 + j], s2[j]);
   }
   __syncthreads();
   for (int cc = threadIdx.x; cc < C; cc += blockDim.x) {
@@ -193,35 +177,34 @@ __global__ void bn_act_bwd_apply_kernel(
     const float* __restrict__ a, const float* __restrict__ sums,
     T* __restrict__ dx, T* __restrict__ dskip, int64_t n_pix, int C,
     float inv_n) {
-  const bool bf16 = sizeof(T) == 2;
-  int c, r_off, rows, per;
-  if (C <= 256) { rows = 256 / C; c = threadIdx.x & (C - 1);
-                  r_off = threadIdx.x / C; per = 1; }
-  else { rows = 1; c = threadIdx.x; r_off = 0; per = C / 256; }
-  for (int j = 0; j < per; ++j) {
-    int cc = c + j * 256;
-    float mu = mean[cc], is = invstd[cc];
-    float gis = gamma[cc] * is;
-    float av = (ACT == 1) ? a[cc] : 0.f;
-    float sdz_n = sums[cc * 3 + 0] * inv_n;
-    float sdzx_n = sums[cc * 3 + 1] * inv_n;
-    for (int64_t p = (int64_t)blockIdx.x * rows + r_off; p < n_pix;
-         p += (int64_t)gridDim.x * rows) {
-      int64_t i = p * C + cc;
-      float dyi = load_f(dy, i, bf16);
-      float dz = dyi;
-      if (ACT == 1) {
-        float zi = load_f(z, i, bf16);
-        dz = zi > 0.f ? dyi : av * dyi;
-      } else if (ACT == 2) {
-        float zi = load_f(z, i, bf16);
-        dz = zi > 0.f ? dyi : 0.f;
-      }
-      float xhat = (load_f(x, i, bf16) - mu) * is;
-      float dxi = gis * (dz - sdz_n - xhat * sdzx_n);
-      store_f(dx, i, dxi, bf16);
-      if (dskip != nullptr) store_f(dskip, i, dz, bf16);
+  ChanMap m = chan_map8(C);
+  float mu[8], is[8], gis[8], av[8], sdz_n[8], sdzx_n[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int cc = m.c0 + j;
+    mu[j] = mean[cc]; is[j] = invstd[cc];
+    gis[j] = gamma[cc] * is[j];
+    av[j] = (ACT == 1) ? a[cc] : 0.f;
+    sdz_n[j] = sums[cc * 3 + 0] * inv_n;
+    sdzx_n[j] = sums[cc * 3 + 1] * inv_n;
+  }
+  float dyv[8], zv[8], xv[8], dxv[8], dzv[8];
+  for (int64_t p = m.p0; p < n_pix; p += m.pstep) {
+    int64_t i = p * C + m.c0;
+    load8(dy, i, dyv);
+    load8(x, i, xv);
+    if (ACT != 0) load8(z, i, zv);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float dz = dyv[j];
+      if (ACT == 1) dz = zv[j] > 0.f ? dyv[j] : av[j] * dyv[j];
+      else if (ACT == 2) dz = zv[j] > 0.f ? dyv[j] : 0.f;
+      float xhat = (xv[j] - mu[j]) * is[j];
+      dxv[j] = gis[j] * (dz - sdz_n[j] - xhat * sdzx_n[j]);
+      dzv[j] = dz;
     }
+    store8(dx, i, dxv);
+    if (dskip != nullptr) store8(dskip, i, dzv);
   }
 }
 
@@ -236,34 +219,34 @@ __global__ void bn_act_eval_kernel(const T* __restrict__ x,
                                    const float* __restrict__ a,
                                    T* __restrict__ out, int64_t n_pix, int C,
                                    float eps) {
-  const bool bf16 = sizeof(T) == 2;
-  int c, r_off, rows, per;
-  if (C <= 256) { rows = 256 / C; c = threadIdx.x & (C - 1);
-                  r_off = threadIdx.x / C; per = 1; }
-  else { rows = 1; c = threadIdx.x; r_off = 0; per = C / 256; }
-  for (int j = 0; j < per; ++j) {
-    int cc = c + j * 256;
+  ChanMap m = chan_map8(C);
+  float sc[8], sh[8], av[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int cc = m.c0 + j;
     float is = rsqrtf(rv[cc] + eps);
-    float sc = gamma[cc] * is;
-    float sh = beta[cc] - rm[cc] * sc;
-    float av = (ACT == 1) ? a[cc] : 0.f;
-    for (int64_t p = (int64_t)blockIdx.x * rows + r_off; p < n_pix;
-         p += (int64_t)gridDim.x * rows) {
-      int64_t i = p * C + cc;
-      float zv = sc * load_f(x, i, bf16) + sh;
-      if (skip != nullptr) zv += load_f(skip, i, bf16);
-      float o = zv;
-      if (ACT == 1) o = zv > 0.f ? zv : av * zv;
-      else if (ACT == 2) o = fmaxf(zv, 0.f);
-      store_f(out, i, o, bf16);
+    sc[j] = gamma[cc] * is;
+    sh[j] = beta[cc] - rm[cc] * sc[j];
+    av[j] = (ACT == 1) ? a[cc] : 0.f;
+  }
+  float v[8], s[8], o[8];
+  for (int64_t p = m.p0; p < n_pix; p += m.pstep) {
+    int64_t i = p * C + m.c0;
+    load8(x, i, v);
+    if (skip != nullptr) load8(skip, i, s);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float zv = sc[j] * v[j] + sh[j];
+      if (skip != nullptr) zv += s[j];
+      o[j] = zv;
+      if (ACT == 1) o[j] = zv > 0.f ? zv : av[j] * zv;
+      else if (ACT == 2) o[j] = fmaxf(zv, 0.f);
     }
+    store8(out, i, o);
   }
 }
 
-static inline int grid_for_pix(int64_t n_pix, int C) {
-  int rows = C <= 256 ? 256 / C : 1;
-  return (int)bd_min<int64_t>((n_pix + rows - 1) / rows, 2048);
-}
+// ---------------- launchers ----------------
 
 template <typename T>
 static void launch_fwd(const void* x, const void* skip, const float* mean,
@@ -346,7 +329,7 @@ extern "C" void bdbnn_bn_stats(const void* x, float* s1, float* s2,
   hipMemsetAsync(s1, 0, sizeof(float) * C, stream);
   hipMemsetAsync(s2, 0, sizeof(float) * C, stream);
   int64_t n_pix = n / C;
-  int grid = grid_for_pix(n_pix, C);
+  int grid = grid_pix8(n_pix, C);
   size_t lds = 2 * sizeof(float) * C;
   if (bf16)
    hipLaunchKernelGGL(( bn_stats_kernel<uint16_t>), dim3(grid), dim3(256), lds, stream, 
@@ -372,7 +355,7 @@ extern "C" void bdbnn_bn_act_fwd(const void* x, const void* skip,
                                  int64_t n, int C, int act_kind, bool bf16,
                                  hipStream_t stream) {
   int64_t n_pix = n / C;
-  int grid = grid_for_pix(n_pix, C);
+  int grid = grid_pix8(n_pix, C);
   if (bf16)
     launch_fwd<uint16_t>(x, skip, mean, invstd, gamma, beta, a, out, zout,
                          n_pix, C, act_kind, grid, stream);
@@ -389,7 +372,7 @@ extern "C" void bdbnn_bn_act_bwd_reduce(const void* dy, const void* z,
                                         hipStream_t stream) {
   hipMemsetAsync(sums, 0, sizeof(float) * C * 3, stream);
   int64_t n_pix = n / C;
-  int grid = grid_for_pix(n_pix, C);
+  int grid = grid_pix8(n_pix, C);
   size_t lds = 3 * sizeof(float) * C;
   if (bf16)
     launch_bwd_reduce<uint16_t>(dy, z, x, mean, invstd, a, sums, n_pix, C,
@@ -408,7 +391,7 @@ extern "C" void bdbnn_bn_act_bwd_apply(const void* dy, const void* z,
                                        int act_kind, float inv_n, bool bf16,
                                        hipStream_t stream) {
   int64_t n_pix = n / C;
-  int grid = grid_for_pix(n_pix, C);
+  int grid = grid_pix8(n_pix, C);
   if (bf16)
     launch_bwd_apply<uint16_t>(dy, z, x, mean, invstd, gamma, a, sums, dx,
                                dskip, n_pix, C, act_kind, inv_n, grid,
@@ -425,7 +408,7 @@ extern "C" void bdbnn_bn_act_eval(const void* x, const void* skip,
                                   int C, int act_kind, float eps, bool bf16,
                                   hipStream_t stream) {
   int64_t n_pix = n / C;
-  int grid = grid_for_pix(n_pix, C);
+  int grid = grid_pix8(n_pix, C);
   if (bf16)
     launch_eval<uint16_t>(x, skip, rm, rv, gamma, beta, a, out, n_pix, C,
                           act_kind, eps, grid, stream);

@@ -10,6 +10,7 @@
 // stores 0, w-pack stores 1 — every garbage bit then contributes exactly 1
 // to the popcount sum, a compile-known constant the epilogue subtracts.
 #include "common.h"
+#include "vec8.h"
 
 // ---------------- activation sign+pack ----------------
 // x: NHWC-contiguous (channels_last torch tensor), P = N*H*W pixels,
@@ -50,16 +51,18 @@ extern "C" void bdbnn_sign_pack(const void* x, uint32_t* out, int64_t pixels,
 }
 
 // ---------------- +-1 decode (for the dense MFMA backward) ----------------
+// 8-wide vectorized (n % 8 == 0 guaranteed by callers: numel is a
+// multiple of C >= 8).
 template <typename TI, typename TO>
 __global__ void binsign_decode_kernel(const TI* __restrict__ x,
-                                      TO* __restrict__ y, int64_t n) {
-  GRID_STRIDE(i, n) {
-    float v;
-    if constexpr (sizeof(TI) == 2) v = bf16_to_f32(((const uint16_t*)x)[i]);
-    else                           v = ((const float*)x)[i];
-    float s = v >= 0.f ? 1.f : -1.f;
-    if constexpr (sizeof(TO) == 2) ((uint16_t*)y)[i] = f32_to_bf16(s);
-    else                           ((float*)y)[i] = s;
+                                      TO* __restrict__ y, int64_t n8) {
+  GRID_STRIDE(q, n8) {
+    int64_t i = q * 8;
+    float v[8], s[8];
+    load8(x, i, v);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) s[j] = v[j] >= 0.f ? 1.f : -1.f;
+    store8(y, i, s);
   }
 }
 
@@ -67,7 +70,9 @@ extern "C" void bdbnn_binsign_decode(const void* x, void* y, int64_t n,
                                      bool in_bf16, bool out_bf16,
                                      hipStream_t stream) {
   int block = 256;
-  int grid = (int)bd_min<int64_t>((n + block - 1) / block, 65535 * 8);
+  int64_t n8 = n / 8;   // callers guarantee n % 8 == 0
+  int grid = (int)bd_min<int64_t>((n8 + block - 1) / block, 2048);
+  n = n8;
 #define CASE(IB, OB, TI, TO)                                              \
   if (in_bf16 == IB && out_bf16 == OB)                                    \
    hipLaunchKernelGGL(( binsign_decode_kernel<TI, TO>), dim3(grid), dim3(block), 0, stream,             \
@@ -85,27 +90,28 @@ extern "C" void bdbnn_binsign_decode(const void* x, void* y, int64_t n,
 template <typename TG, typename TX>
 __global__ void ste_mask_mul_kernel(const TG* __restrict__ g,
                                     const TX* __restrict__ x,
-                                    TX* __restrict__ y, int64_t n,
+                                    TX* __restrict__ y, int64_t n8,
                                     int mode, float t, float k) {
-  GRID_STRIDE(i, n) {
-    float gv, xv;
-    if constexpr (sizeof(TG) == 2) gv = bf16_to_f32(((const uint16_t*)g)[i]);
-    else                           gv = ((const float*)g)[i];
-    if constexpr (sizeof(TX) == 2) xv = bf16_to_f32(((const uint16_t*)x)[i]);
-    else                           xv = ((const float*)x)[i];
-    float m;
-    if (mode == 0) {
-      m = fabsf(xv) <= 1.f ? 1.f : 0.f;
-    } else if (mode == 1) {
-      m = (xv >= -1.f && xv < 0.f) ? 2.f + 2.f * xv
-        : (xv >= 0.f && xv < 1.f) ? 2.f - 2.f * xv : 0.f;
-    } else {
-      float th = tanhf(t * xv);
-      m = k * t * (1.f - th * th);
+  GRID_STRIDE(q, n8) {
+    int64_t i = q * 8;
+    float gv[8], xv[8], o[8];
+    load8(g, i, gv);
+    load8(x, i, xv);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float m;
+      if (mode == 0) {
+        m = fabsf(xv[j]) <= 1.f ? 1.f : 0.f;
+      } else if (mode == 1) {
+        m = (xv[j] >= -1.f && xv[j] < 0.f) ? 2.f + 2.f * xv[j]
+          : (xv[j] >= 0.f && xv[j] < 1.f) ? 2.f - 2.f * xv[j] : 0.f;
+      } else {
+        float th = tanhf(t * xv[j]);
+        m = k * t * (1.f - th * th);
+      }
+      o[j] = gv[j] * m;
     }
-    float out = gv * m;
-    if constexpr (sizeof(TX) == 2) ((uint16_t*)y)[i] = f32_to_bf16(out);
-    else                           ((float*)y)[i] = out;
+    store8(y, i, o);
   }
 }
 
@@ -114,7 +120,9 @@ extern "C" void bdbnn_ste_mask_mul(const void* g, const void* x, void* y,
                                    int mode, float t, float k,
                                    hipStream_t stream) {
   int block = 256;
-  int grid = (int)bd_min<int64_t>((n + block - 1) / block, 65535 * 8);
+  int64_t n8 = n / 8;   // callers guarantee n % 8 == 0
+  int grid = (int)bd_min<int64_t>((n8 + block - 1) / block, 2048);
+  n = n8;
 #define CASE(GB, XB, TG, TX)                                              \
   if (g_bf16 == GB && x_bf16 == XB)                                       \
    hipLaunchKernelGGL(( ste_mask_mul_kernel<TG, TX>), dim3(grid), dim3(block), 0, stream,               \
