@@ -1,0 +1,93 @@
+"""Binary inference path (BASELINE config 5): persistent packed 1-bit
+weights + hipGraph-captured eval loop.
+
+``PackedInference(model)`` snapshots every HardBinaryConv's packed
+weights once (bit-pack + alpha + pad table survive across calls instead
+of being recomputed per forward), switches the model to eval, and
+``capture(batch_shape)`` records the whole forward into a hipGraph so
+steady-state inference replays with near-zero launch overhead
+(SURVEY.md K11).
+"""
+
+import torch
+
+from .. import _C
+from ..ops.binary_conv import _HardBinaryConvBase
+
+
+class _PackedConvForward:
+    """Replacement forward for HardBinaryConv in inference: cached packed
+    weights, no autograd, no repack."""
+
+    def __init__(self, conv):
+        nat = _C.native_required()
+        self.stride = conv.stride
+        self.padding = conv.padding
+        self.C = conv.in_channels
+        self.wp, self.alpha, self.stab = nat.weight_pack(conv.weight.detach())
+
+    def __call__(self, x):
+        nat = _C.native_required()
+        xc = x.contiguous(memory_format=torch.channels_last)
+        xp = nat.sign_pack_nhwc(xc)
+        return nat.xnor_conv_fwd(xp, self.wp, self.alpha, self.stab,
+                                 self.C, self.stride, self.padding,
+                                 x.dtype == torch.bfloat16)
+
+
+class PackedInference:
+    """Inference wrapper: packed weights resident in HBM + hipGraph replay."""
+
+    def __init__(self, model, dtype=torch.bfloat16):
+        assert torch.cuda.is_available(), "PackedInference needs a GPU"
+        _C.native_required()
+        self.dtype = dtype
+        self.model = model.cuda().to(memory_format=torch.channels_last).eval()
+        self._orig_forwards = {}
+        for mod in self.model.modules():
+            if isinstance(mod, _HardBinaryConvBase):
+                packed = _PackedConvForward(mod)
+                self._orig_forwards[mod] = mod.forward
+                mod.forward = packed
+        self.graph = None
+        self._static_in = None
+        self._static_out = None
+
+    @torch.no_grad()
+    def __call__(self, x):
+        x = x.to("cuda", self.dtype).contiguous(
+            memory_format=torch.channels_last)
+        if self.graph is not None and x.shape == self._static_in.shape:
+            self._static_in.copy_(x)
+            self.graph.replay()
+            return self._static_out
+        with torch.autocast("cuda", dtype=torch.bfloat16,
+                            enabled=self.dtype == torch.bfloat16):
+            return self.model(x)
+
+    @torch.no_grad()
+    def capture(self, batch_shape, warmup=3):
+        """Record one forward of the given shape into a hipGraph."""
+        self._static_in = torch.zeros(
+            batch_shape, device="cuda", dtype=self.dtype).contiguous(
+            memory_format=torch.channels_last)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(warmup):
+                with torch.autocast("cuda", dtype=torch.bfloat16,
+                                    enabled=self.dtype == torch.bfloat16):
+                    out = self.model(self._static_in)
+        torch.cuda.current_stream().wait_stream(s)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            with torch.autocast("cuda", dtype=torch.bfloat16,
+                                enabled=self.dtype == torch.bfloat16):
+                self._static_out = self.model(self._static_in)
+        return self
+
+    def release(self):
+        for mod, fwd in self._orig_forwards.items():
+            mod.forward = fwd
+        self._orig_forwards.clear()
+        self.graph = None

@@ -32,6 +32,9 @@ def parse_args():
     p.add_argument("--arch", default="resnet18")
     p.add_argument("--ts", action="store_true",
                    help="teacher-student KD path (BASELINE config 3)")
+    p.add_argument("--infer", action="store_true",
+                   help="binary inference path: packed weights + hipGraph "
+                        "(BASELINE config 5, throughput-only)")
     p.add_argument("--no-kurt", action="store_true")
     p.add_argument("--image", type=int, default=224)
     return p.parse_args()
@@ -77,6 +80,36 @@ def main():
         start_epoch = 0
 
     model = im.__dict__[args.arch](False)
+
+    if args.infer:
+        from bdbnn_amd.engine import PackedInference
+        assert use_cuda, "--infer needs a GPU"
+        eng = PackedInference(model).capture(
+            (args.batch_size, 3, args.image, args.image))
+        x = torch.randn(args.batch_size, 3, args.image, args.image,
+                        device=device)
+        for _ in range(args.warmup):
+            eng(x)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(args.steps):
+            eng(x)
+        torch.cuda.synchronize()
+        elapsed = time.perf_counter() - t0
+        print(json.dumps({
+            "metric": "infer_images_per_sec",
+            "value": round(args.batch_size * args.steps / elapsed, 2),
+            "unit": "images/s", "n_gpus": 1, "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000.0, 3),
+            "higher_is_better": True, "scaling": "weak",
+            "vs_baseline": None, "dtype": "bf16", "data": "synthetic",
+            "config": {"model": f"{args.arch}_bdbnn_infer",
+                       "global_batch": args.batch_size,
+                       "image": f"3x{args.image}x{args.image}",
+                       "parallelism": "dp1", "hipgraph": True}}))
+        return
+
     teacher = im.__dict__[args.arch + "_real"](False) if args.ts else None
     trainer = Trainer(model, TArgs, teacher=teacher, device=device,
                       world_size=world_size, rank=rank)

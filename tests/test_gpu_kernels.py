@@ -342,3 +342,39 @@ def test_fused_bn_act_eval():
         out = fused_bn_act(x, bn, "relu")
         ref = torch.relu(bn(x))
     assert torch.allclose(out, _cl(ref), atol=1e-4, rtol=1e-4)
+
+
+# ---------------- inference path ----------------
+
+def test_packed_inference_matches_module_forward():
+    from bdbnn_amd.models import imagenet as im
+    from bdbnn_amd.engine import PackedInference
+    torch.manual_seed(12)
+    model = im.resnet18(False)
+    ref = im.resnet18(False)
+    ref.load_state_dict(model.state_dict())
+    ref = ref.cuda().to(memory_format=torch.channels_last).eval()
+    eng = PackedInference(model, dtype=torch.float32)
+    x = torch.randn(4, 3, 64, 64, device="cuda")
+    out = eng(x)
+    with torch.no_grad():
+        want = ref(_cl(x))
+    assert torch.allclose(out, want, atol=2e-2, rtol=1e-2), \
+        (out - want).abs().max().item()
+
+
+def test_packed_inference_hipgraph_replay():
+    from bdbnn_amd.models import imagenet as im
+    from bdbnn_amd.engine import PackedInference
+    torch.manual_seed(13)
+    model = im.resnet18(False)
+    eng = PackedInference(model, dtype=torch.float32)
+    x = torch.randn(2, 3, 64, 64, device="cuda")
+    eager = eng(x).clone()
+    eng.capture((2, 3, 64, 64))
+    replayed = eng(x).clone()
+    assert torch.allclose(eager, replayed, atol=1e-3, rtol=1e-3)
+    # replay twice with different inputs gives different outputs
+    x2 = torch.randn(2, 3, 64, 64, device="cuda")
+    out2 = eng(x2).clone()
+    assert not torch.allclose(replayed, out2)
