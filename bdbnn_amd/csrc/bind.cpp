@@ -28,6 +28,14 @@ void bdbnn_weight_pack(const float*, uint32_t*, float*, float*, int, int,
 void bdbnn_xnor_conv_fwd(const uint32_t*, const uint32_t*, const float*,
                          const float*, void*, bool, int, int, int, int, int,
                          int, int, int, int, int, int, hipStream_t);
+void bdbnn_sign_mask_pack(const void*, uint32_t*, uint32_t*, int64_t, int,
+                          int, bool, hipStream_t);
+void bdbnn_decode_packed(const uint32_t*, void*, int64_t, int, int, bool,
+                         hipStream_t);
+void bdbnn_mask_mul_packed(const void*, const uint32_t*, void*, int64_t,
+                           int, int, bool, bool, hipStream_t);
+void bdbnn_weight_decode(const uint32_t*, const float*, void*, int, int,
+                         int, int, bool, hipStream_t);
 void bdbnn_prelu_fwd(const void*, const float*, void*, int64_t, int, bool,
                      hipStream_t);
 void bdbnn_prelu_bwd(const void*, const void*, const float*, void*, float*,
@@ -189,6 +197,68 @@ at::Tensor ste_mask_mul(const at::Tensor& g, const at::Tensor& x, int mode,
   bdbnn_ste_mask_mul(gc.data_ptr(), xc.data_ptr(), out.data_ptr(),
                      xc.numel(), is_bf16(gc), is_bf16(xc), mode, (float)t,
                      (float)k, cur_stream());
+  return out;
+}
+
+// ---------------- packed-bit fast path ----------------
+
+std::vector<at::Tensor> sign_mask_pack_nhwc(const at::Tensor& x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4, "sign_mask_pack: 4-D CUDA tensor");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "sign_mask_pack: channels_last input required");
+  int64_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int CW = (int)((C + 31) / 32);
+  auto sp = at::empty({N, H, W, CW}, x.options().dtype(at::kInt));
+  auto mp = at::empty({N, H, W, CW}, x.options().dtype(at::kInt));
+  bdbnn_sign_mask_pack(x.data_ptr(), (uint32_t*)sp.data_ptr<int>(),
+                       (uint32_t*)mp.data_ptr<int>(), N * H * W, (int)C, CW,
+                       is_bf16(x), cur_stream());
+  return {sp, mp};
+}
+
+at::Tensor decode_packed(const at::Tensor& sp, int64_t C, bool out_bf16) {
+  TORCH_CHECK(sp.is_cuda() && sp.dim() == 4 && sp.scalar_type() == at::kInt,
+              "decode_packed: int32 [N,H,W,CW]");
+  int64_t N = sp.size(0), H = sp.size(1), W = sp.size(2);
+  int CW = (int)sp.size(3);
+  auto out = at::empty({N, C, H, W},
+                       sp.options().dtype(out_bf16 ? at::kBFloat16
+                                                   : at::kFloat),
+                       at::MemoryFormat::ChannelsLast);
+  bdbnn_decode_packed((const uint32_t*)sp.data_ptr<int>(), out.data_ptr(),
+                      N * H * W, (int)C, CW, out_bf16, cur_stream());
+  return out;
+}
+
+at::Tensor mask_mul_packed(const at::Tensor& g, const at::Tensor& mp,
+                           int64_t C, bool out_bf16) {
+  TORCH_CHECK(g.is_cuda() && mp.is_cuda(), "mask_mul_packed: CUDA tensors");
+  auto gc = g.contiguous(at::MemoryFormat::ChannelsLast);
+  int64_t N = mp.size(0), H = mp.size(1), W = mp.size(2);
+  int CW = (int)mp.size(3);
+  TORCH_CHECK(gc.numel() == N * H * W * C, "mask_mul_packed: shape mismatch");
+  auto out = at::empty({N, C, H, W},
+                       g.options().dtype(out_bf16 ? at::kBFloat16
+                                                  : at::kFloat),
+                       at::MemoryFormat::ChannelsLast);
+  bdbnn_mask_mul_packed(gc.data_ptr(), (const uint32_t*)mp.data_ptr<int>(),
+                        out.data_ptr(), N * H * W, (int)C, CW, is_bf16(gc),
+                        out_bf16, cur_stream());
+  return out;
+}
+
+at::Tensor weight_decode(const at::Tensor& wp, const at::Tensor& alpha,
+                         int64_t C, bool out_bf16) {
+  TORCH_CHECK(wp.is_cuda() && wp.dim() == 4 && wp.scalar_type() == at::kInt,
+              "weight_decode: int32 [K,KH,KW,CW]");
+  int K = (int)wp.size(0), KH = (int)wp.size(1), KW = (int)wp.size(2);
+  int CW = (int)wp.size(3);
+  auto out = at::empty({K, C, KH, KW},
+                       wp.options().dtype(out_bf16 ? at::kBFloat16
+                                                   : at::kFloat));
+  bdbnn_weight_decode((const uint32_t*)wp.data_ptr<int>(),
+                      alpha.data_ptr<float>(), out.data_ptr(), K, (int)C,
+                      KH * KW, CW, out_bf16, cur_stream());
   return out;
 }
 
@@ -489,6 +559,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "pack conv weights -> (bits, alpha, pad-correction table)");
   m.def("binsign_decode", &binsign_decode, "elementwise +-1 decode");
   m.def("ste_mask_mul", &ste_mask_mul, "quantizer backward mask multiply");
+  m.def("sign_mask_pack_nhwc", &sign_mask_pack_nhwc,
+        "pack sign + clip-STE mask bitplanes in one pass");
+  m.def("decode_packed", &decode_packed, "packed signs -> +-1 NHWC tensor");
+  m.def("mask_mul_packed", &mask_mul_packed, "dx = mask_bit ? g : 0");
+  m.def("weight_decode", &weight_decode, "packed weights -> alpha*(+-1)");
   m.def("xnor_conv_fwd", &xnor_conv_fwd, "bit-packed XNOR+popcount conv");
   m.def("prelu_fwd", &prelu_fwd, "fused NHWC per-channel PReLU fwd");
   m.def("prelu_bwd", &prelu_bwd, "fused NHWC per-channel PReLU bwd");

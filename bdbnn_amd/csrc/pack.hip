@@ -196,3 +196,163 @@ extern "C" void bdbnn_weight_pack(const float* w, uint32_t* wp, float* alpha,
   hipMemsetAsync(stab, 0, sizeof(float) * K * KH * KW, stream);
   weight_pack_kernel<<<grid, block, 0, stream>>>(w, wp, stab, K, C, KH, KW, CW);
 }
+
+// ---------------- sign+mask pack (one pass) ----------------
+// Emits BOTH bitplanes the 'ste' backward needs: sign bits (for the
+// XNOR conv and the +-1 decode) and |x| <= 1 clip-STE mask bits.
+// Backward then never touches the fp activations, and the forward does
+// not save them.
+template <typename T>
+__global__ void sign_mask_pack_kernel(const T* __restrict__ x,
+                                      uint32_t* __restrict__ sp,
+                                      uint32_t* __restrict__ mp,
+                                      int64_t n_words, int C, int CW) {
+  GRID_STRIDE(i, n_words) {
+    int cw = int(i % CW);
+    int64_t p = i / CW;
+    const T* px = x + p * C + cw * 32;
+    int nbits = min(32, C - cw * 32);
+    uint32_t sbits = 0, mbits = 0;
+#pragma unroll 8
+    for (int c = 0; c < nbits; ++c) {
+      float v;
+      if constexpr (sizeof(T) == 2) v = bf16_to_f32(((const uint16_t*)px)[c]);
+      else                          v = ((const float*)px)[c];
+      sbits |= (v >= 0.f ? 1u : 0u) << c;
+      mbits |= (fabsf(v) <= 1.f ? 1u : 0u) << c;
+    }
+    sp[i] = sbits;
+    mp[i] = mbits;
+  }
+}
+
+extern "C" void bdbnn_sign_mask_pack(const void* x, uint32_t* sp,
+                                     uint32_t* mp, int64_t pixels, int C,
+                                     int CW, bool bf16, hipStream_t stream) {
+  int64_t n_words = pixels * CW;
+  int block = 256;
+  int grid = (int)bd_min<int64_t>((n_words + block - 1) / block, 4096);
+  if (bf16)
+    sign_mask_pack_kernel<uint16_t><<<grid, block, 0, stream>>>(
+        (const uint16_t*)x, sp, mp, n_words, C, CW);
+  else
+    sign_mask_pack_kernel<float><<<grid, block, 0, stream>>>(
+        (const float*)x, sp, mp, n_words, C, CW);
+}
+
+// ---------------- decode packed signs -> +-1 NHWC tensor ----------------
+// one thread per 32-channel word; writes 32 consecutive elements.
+template <typename TO>
+__global__ void decode_packed_kernel(const uint32_t* __restrict__ sp,
+                                     TO* __restrict__ y, int64_t n_words,
+                                     int C, int CW) {
+  GRID_STRIDE(i, n_words) {
+    int cw = int(i % CW);
+    int64_t p = i / CW;
+    int nbits = min(32, C - cw * 32);
+    uint32_t bits = sp[i];
+    TO* py = y + p * C + cw * 32;
+#pragma unroll 8
+    for (int c = 0; c < nbits; ++c) {
+      float s = (bits >> c) & 1 ? 1.f : -1.f;
+      if constexpr (sizeof(TO) == 2) ((uint16_t*)py)[c] = f32_to_bf16(s);
+      else                           ((float*)py)[c] = s;
+    }
+  }
+}
+
+extern "C" void bdbnn_decode_packed(const uint32_t* sp, void* y,
+                                    int64_t pixels, int C, int CW,
+                                    bool out_bf16, hipStream_t stream) {
+  int64_t n_words = pixels * CW;
+  int block = 256;
+  int grid = (int)bd_min<int64_t>((n_words + block - 1) / block, 4096);
+  if (out_bf16)
+    decode_packed_kernel<uint16_t><<<grid, block, 0, stream>>>(
+        sp, (uint16_t*)y, n_words, C, CW);
+  else
+    decode_packed_kernel<float><<<grid, block, 0, stream>>>(
+        sp, (float*)y, n_words, C, CW);
+}
+
+// ---------------- masked grad: dx = mask_bit ? g : 0 ----------------
+template <typename TG, typename TO>
+__global__ void mask_mul_packed_kernel(const TG* __restrict__ g,
+                                       const uint32_t* __restrict__ mp,
+                                       TO* __restrict__ dx, int64_t n_words,
+                                       int C, int CW) {
+  GRID_STRIDE(i, n_words) {
+    int cw = int(i % CW);
+    int64_t p = i / CW;
+    int nbits = min(32, C - cw * 32);
+    uint32_t bits = mp[i];
+    const TG* pg = g + p * C + cw * 32;
+    TO* pd = dx + p * C + cw * 32;
+#pragma unroll 8
+    for (int c = 0; c < nbits; ++c) {
+      float gv;
+      if constexpr (sizeof(TG) == 2) gv = bf16_to_f32(((const uint16_t*)pg)[c]);
+      else                           gv = ((const float*)pg)[c];
+      float o = (bits >> c) & 1 ? gv : 0.f;
+      if constexpr (sizeof(TO) == 2) ((uint16_t*)pd)[c] = f32_to_bf16(o);
+      else                           ((float*)pd)[c] = o;
+    }
+  }
+}
+
+extern "C" void bdbnn_mask_mul_packed(const void* g, const uint32_t* mp,
+                                      void* dx, int64_t pixels, int C,
+                                      int CW, bool g_bf16, bool out_bf16,
+                                      hipStream_t stream) {
+  int64_t n_words = pixels * CW;
+  int block = 256;
+  int grid = (int)bd_min<int64_t>((n_words + block - 1) / block, 4096);
+#define MCASE(GB, OB, TG, TO)                                             \
+  if (g_bf16 == GB && out_bf16 == OB)                                     \
+    mask_mul_packed_kernel<TG, TO><<<grid, block, 0, stream>>>(           \
+        (const TG*)g, mp, (TO*)dx, n_words, C, CW);
+  MCASE(false, false, float, float)
+  MCASE(false, true, float, uint16_t)
+  MCASE(true, false, uint16_t, float)
+  MCASE(true, true, uint16_t, uint16_t)
+#undef MCASE
+}
+
+// ---------------- decode packed weights -> alpha * (+-1), NCHW ----------
+// wp: [K][KH][KW][CW] inverted-bit convention (bit 1 <=> w < 0).
+template <typename TO>
+__global__ void weight_decode_kernel(const uint32_t* __restrict__ wp,
+                                     const float* __restrict__ alpha,
+                                     TO* __restrict__ w, int K, int C,
+                                     int T, int CW) {
+  int64_t n_words = (int64_t)K * T * CW;
+  GRID_STRIDE(i, n_words) {
+    int cw = int(i % CW);
+    int t = int((i / CW) % T);
+    int k = int(i / CW / T);
+    int nbits = min(32, C - cw * 32);
+    uint32_t bits = wp[i];
+    float al = alpha[k];
+    // w layout [K][C][KH][KW]: stride of c is T
+    TO* pw = w + ((int64_t)k * C + cw * 32) * T + t;
+    for (int c = 0; c < nbits; ++c) {
+      float v = (bits >> c) & 1 ? -al : al;  // inverted convention
+      if constexpr (sizeof(TO) == 2) ((uint16_t*)pw)[(int64_t)c * T] = f32_to_bf16(v);
+      else                           ((float*)pw)[(int64_t)c * T] = v;
+    }
+  }
+}
+
+extern "C" void bdbnn_weight_decode(const uint32_t* wp, const float* alpha,
+                                    void* w, int K, int C, int T, int CW,
+                                    bool out_bf16, hipStream_t stream) {
+  int64_t n_words = (int64_t)K * T * CW;
+  int block = 256;
+  int grid = (int)bd_min<int64_t>((n_words + block - 1) / block, 4096);
+  if (out_bf16)
+    weight_decode_kernel<uint16_t><<<grid, block, 0, stream>>>(
+        wp, alpha, (uint16_t*)w, K, C, T, CW);
+  else
+    weight_decode_kernel<float><<<grid, block, 0, stream>>>(
+        wp, alpha, (float*)w, K, C, T, CW);
+}

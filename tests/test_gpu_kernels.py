@@ -378,3 +378,30 @@ def test_packed_inference_hipgraph_replay():
     x2 = torch.randn(2, 3, 64, 64, device="cuda")
     out2 = eng(x2).clone()
     assert not torch.allclose(replayed, out2)
+
+
+# ---------------- packed-bit backward path ----------------
+
+def test_sign_mask_pack_and_decode():
+    torch.manual_seed(14)
+    x = torch.randn(2, 64, 6, 6, device="cuda") * 2
+    nat = _nat()
+    sp, mp = nat.sign_mask_pack_nhwc(_cl(x))
+    sp2 = nat.sign_pack_nhwc(_cl(x))
+    assert torch.equal(sp, sp2)
+    xb = nat.decode_packed(sp, 64, False)
+    assert torch.equal(xb, _cl(binsign(x)))
+    g = torch.randn(2, 64, 6, 6, device="cuda")
+    dx = nat.mask_mul_packed(_cl(g), mp, 64, False)
+    ref = _cl(g * (x.abs() <= 1).float())
+    assert torch.equal(dx, ref)
+
+
+def test_weight_decode_roundtrip():
+    torch.manual_seed(15)
+    w = torch.randn(32, 48, 3, 3, device="cuda")  # tail-word C
+    nat = _nat()
+    wp, alpha, stab = nat.weight_pack(w)
+    wb = nat.weight_decode(wp, alpha, 48, False)
+    ref = weight_scale(w) * binsign(w)
+    assert torch.allclose(wb, ref, atol=1e-6)
