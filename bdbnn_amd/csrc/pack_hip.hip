@@ -25,12 +25,21 @@ __global__ void sign_pack_kernel(const T* __restrict__ x,
     const T* px = x + p * C + cw * 32;
     int nbits = min(32, C - cw * 32);
     uint32_t bits = 0;
-#pragma unroll 8
-    for (int c = 0; c < nbits; ++c) {
-      float v;
-      if constexpr (sizeof(T) == 2) v = bf16_to_f32(((const uint16_t*)px)[c]);
-      else                          v = ((const float*)px)[c];
-      bits |= (v >= 0.f ? 1u : 0u) << c;
+    if (nbits == 32) {
+#pragma unroll
+      for (int c = 0; c < 32; ++c) {
+        float v;
+        if constexpr (sizeof(T) == 2) v = bf16_to_f32(((const uint16_t*)px)[c]);
+        else                          v = ((const float*)px)[c];
+        bits |= (v >= 0.f ? 1u : 0u) << c;
+      }
+    } else {
+      for (int c = 0; c < nbits; ++c) {
+        float v;
+        if constexpr (sizeof(T) == 2) v = bf16_to_f32(((const uint16_t*)px)[c]);
+        else                          v = ((const float*)px)[c];
+        bits |= (v >= 0.f ? 1u : 0u) << c;
+      }
     }
     out[i] = bits;
   }
@@ -214,13 +223,23 @@ __global__ void sign_mask_pack_kernel(const T* __restrict__ x,
     const T* px = x + p * C + cw * 32;
     int nbits = min(32, C - cw * 32);
     uint32_t sbits = 0, mbits = 0;
-#pragma unroll 8
-    for (int c = 0; c < nbits; ++c) {
-      float v;
-      if constexpr (sizeof(T) == 2) v = bf16_to_f32(((const uint16_t*)px)[c]);
-      else                          v = ((const float*)px)[c];
-      sbits |= (v >= 0.f ? 1u : 0u) << c;
-      mbits |= (fabsf(v) <= 1.f ? 1u : 0u) << c;
+    if (nbits == 32) {
+#pragma unroll
+      for (int c = 0; c < 32; ++c) {
+        float v;
+        if constexpr (sizeof(T) == 2) v = bf16_to_f32(((const uint16_t*)px)[c]);
+        else                          v = ((const float*)px)[c];
+        sbits |= (v >= 0.f ? 1u : 0u) << c;
+        mbits |= (fabsf(v) <= 1.f ? 1u : 0u) << c;
+      }
+    } else {
+      for (int c = 0; c < nbits; ++c) {
+        float v;
+        if constexpr (sizeof(T) == 2) v = bf16_to_f32(((const uint16_t*)px)[c]);
+        else                          v = ((const float*)px)[c];
+        sbits |= (v >= 0.f ? 1u : 0u) << c;
+        mbits |= (fabsf(v) <= 1.f ? 1u : 0u) << c;
+      }
     }
     sp[i] = sbits;
     mp[i] = mbits;
@@ -253,11 +272,19 @@ __global__ void decode_packed_kernel(const uint32_t* __restrict__ sp,
     int nbits = min(32, C - cw * 32);
     uint32_t bits = sp[i];
     TO* py = y + p * C + cw * 32;
-#pragma unroll 8
-    for (int c = 0; c < nbits; ++c) {
-      float s = (bits >> c) & 1 ? 1.f : -1.f;
-      if constexpr (sizeof(TO) == 2) ((uint16_t*)py)[c] = f32_to_bf16(s);
-      else                           ((float*)py)[c] = s;
+    if (nbits == 32) {
+#pragma unroll
+      for (int c = 0; c < 32; ++c) {
+        float s = (bits >> c) & 1 ? 1.f : -1.f;
+        if constexpr (sizeof(TO) == 2) ((uint16_t*)py)[c] = f32_to_bf16(s);
+        else                           ((float*)py)[c] = s;
+      }
+    } else {
+      for (int c = 0; c < nbits; ++c) {
+        float s = (bits >> c) & 1 ? 1.f : -1.f;
+        if constexpr (sizeof(TO) == 2) ((uint16_t*)py)[c] = f32_to_bf16(s);
+        else                           ((float*)py)[c] = s;
+      }
     }
   }
 }
@@ -289,14 +316,25 @@ __global__ void mask_mul_packed_kernel(const TG* __restrict__ g,
     uint32_t bits = mp[i];
     const TG* pg = g + p * C + cw * 32;
     TO* pd = dx + p * C + cw * 32;
-#pragma unroll 8
-    for (int c = 0; c < nbits; ++c) {
-      float gv;
-      if constexpr (sizeof(TG) == 2) gv = bf16_to_f32(((const uint16_t*)pg)[c]);
-      else                           gv = ((const float*)pg)[c];
-      float o = (bits >> c) & 1 ? gv : 0.f;
-      if constexpr (sizeof(TO) == 2) ((uint16_t*)pd)[c] = f32_to_bf16(o);
-      else                           ((float*)pd)[c] = o;
+    if (nbits == 32) {
+#pragma unroll
+      for (int c = 0; c < 32; ++c) {
+        float gv;
+        if constexpr (sizeof(TG) == 2) gv = bf16_to_f32(((const uint16_t*)pg)[c]);
+        else                           gv = ((const float*)pg)[c];
+        float o = (bits >> c) & 1 ? gv : 0.f;
+        if constexpr (sizeof(TO) == 2) ((uint16_t*)pd)[c] = f32_to_bf16(o);
+        else                           ((float*)pd)[c] = o;
+      }
+    } else {
+      for (int c = 0; c < nbits; ++c) {
+        float gv;
+        if constexpr (sizeof(TG) == 2) gv = bf16_to_f32(((const uint16_t*)pg)[c]);
+        else                           gv = ((const float*)pg)[c];
+        float o = (bits >> c) & 1 ? gv : 0.f;
+        if constexpr (sizeof(TO) == 2) ((uint16_t*)pd)[c] = f32_to_bf16(o);
+        else                           ((float*)pd)[c] = o;
+      }
     }
   }
 }

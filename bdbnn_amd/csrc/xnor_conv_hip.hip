@@ -20,13 +20,14 @@
 //   dot(sp,k) = 2*POP + BASE - sum_{t invalid(sp)} S[k][t],
 //   BASE = -2*G - C*T,  G = (32*CW - C) * T.
 //
-// Tiling: 256-thread block computes a 64(spatial) x 64(channel) tile;
-// both operands staged in LDS in 8-word chunks; each thread owns a 4x4
-// register tile read via ds_read_b128 (a4 broadcast within a 16-lane
-// group, b4 conflict-free across the 256-B bank row).
+// Tiling: 256-thread block computes a 128(spatial) x 64(channel) tile;
+// both operands staged in LDS in 8-word chunks; each thread owns an
+// 8(spatial) x 4(channel) register tile read via ds_read_b128 (a-words
+// broadcast within a 16-lane group, b4 conflict-free across the 256-B
+// bank row): 1024 binary MACs per ~70 instructions.
 #include "common.h"
 
-#define TILE_M 64
+#define TILE_M 128
 #define TILE_K 64
 #define CHUNK 8
 
@@ -84,17 +85,17 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
   }
   __syncthreads();
 
-  // thread's register tile: rows r0..r0+3, channels k0..k0+3
-  const int r0 = (tid / 16) * 4;
+  // thread's register tile: rows r0..r0+7, channels kq..kq+3
+  const int r0 = (tid / 16) * 8;
   const int kq = (tid % 16) * 4;
-  int acc[4][4] = {};
+  int acc[8][4] = {};
 
   const int n_chunks = (p.WORDS + CHUNK - 1) / CHUNK;
   for (int ch = 0; ch < n_chunks; ++ch) {
     const int w0 = ch * CHUNK;
-    // ---- stage: 2 a-words + 2 w-words per thread ----
+    // ---- stage: 4 a-words + 2 w-words per thread ----
 #pragma unroll
-    for (int it = 0; it < 2; ++it) {
+    for (int it = 0; it < 4; ++it) {
       int j = tid + it * 256;
       int c = j / TILE_M, r = j % TILE_M;
       int widx = w0 + c;
@@ -108,7 +109,12 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
         }
       }
       a_lds[c][r] = av;
-      int kk = j % TILE_K;  // (TILE_K == TILE_M, same decomposition)
+    }
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int j = tid + it * 256;
+      int c = j / TILE_K, kk = j % TILE_K;
+      int widx = w0 + c;
       int kg = k0_blk + kk;
       uint32_t wv = 0;
       if (widx < p.WORDS && kg < p.K)
@@ -119,14 +125,15 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
     // ---- compute ----
 #pragma unroll
     for (int c = 0; c < CHUNK; ++c) {
-      uint32_t a4[4], b4[4];
-      *(uint4*)a4 = *(const uint4*)&a_lds[c][r0];
+      uint32_t a8[8], b4[4];
+      *(uint4*)&a8[0] = *(const uint4*)&a_lds[c][r0];
+      *(uint4*)&a8[4] = *(const uint4*)&a_lds[c][r0 + 4];
       *(uint4*)b4 = *(const uint4*)&w_lds[c][kq];
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
+      for (int i = 0; i < 8; ++i)
 #pragma unroll
         for (int j = 0; j < 4; ++j)
-          acc[i][j] += __popc(a4[i] ^ b4[j]);
+          acc[i][j] += __popc(a8[i] ^ b4[j]);
     }
     __syncthreads();
   }
@@ -139,7 +146,7 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
     al[j] = (kg < p.K) ? alpha[kg] : 0.f;
   }
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
+  for (int i = 0; i < 8; ++i) {
     int64_t sp = m0 + r0 + i;
     if (sp >= M) continue;
     unsigned inv = row_inv[r0 + i];
