@@ -55,7 +55,10 @@ class FusedSGD(torch.optim.Optimizer):
                 continue
             if ps[0].is_cuda and _C.has_native():
                 gs = [_match_layout(g.float(), p) for p, g in zip(ps, gs)]
-                _C.native_required().fused_sgd(ps, gs, bufs, lr, mom, wd)
+                nat = _C.native_required()
+                for i in range(0, len(ps), 64):  # kernel-arg tensor cap
+                    nat.fused_sgd(ps[i:i + 64], gs[i:i + 64],
+                                  bufs[i:i + 64], lr, mom, wd)
             else:
                 for p, g, b in zip(ps, gs, bufs):
                     if wd != 0:
@@ -103,8 +106,11 @@ class FusedAdam(torch.optim.Optimizer):
             bc2 = 1 - beta2 ** step_t
             if ps[0].is_cuda and _C.has_native():
                 gs = [_match_layout(g.float(), p) for p, g in zip(ps, gs)]
-                _C.native_required().fused_adam(
-                    ps, gs, m1s, m2s, lr, beta1, beta2, eps, wd, bc1, bc2)
+                nat = _C.native_required()
+                for i in range(0, len(ps), 64):  # kernel-arg tensor cap
+                    nat.fused_adam(ps[i:i + 64], gs[i:i + 64],
+                                   m1s[i:i + 64], m2s[i:i + 64],
+                                   lr, beta1, beta2, eps, wd, bc1, bc2)
             else:
                 for p, g, m, v in zip(ps, gs, m1s, m2s):
                     if wd != 0:
