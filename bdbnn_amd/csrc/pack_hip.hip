@@ -171,27 +171,28 @@ __global__ void weight_pack_kernel(const float* __restrict__ w,
                                    uint32_t* __restrict__ wp,
                                    float* __restrict__ stab,
                                    int K, int C, int KH, int KW, int CW) {
-  // one thread per packed word (k, t, cw); w layout stride: w[k][c][kh][kw]
+  // one thread per (k, tap): packs that tap's CW words AND computes the
+  // pad-correction entry S[k][t] = C - 2*popc_real with no atomics.
   int T = KH * KW;
-  int64_t n_words = (int64_t)K * T * CW;
-  GRID_STRIDE(i, n_words) {
-    int cw = int(i % CW);
-    int t = int((i / CW) % T);
-    int k = int(i / CW / T);
-    int nbits = min(32, C - cw * 32);
+  int64_t n_taps = (int64_t)K * T;
+  GRID_STRIDE(i, n_taps) {
+    int t = int(i % T);
+    int k = int(i / T);
     const float* wk = w + ((int64_t)k * C) * T + t;  // + c*T steps
-    uint32_t bits = 0;
-    int pop_real = 0;  // popc of REAL (non-inverted) sign bits
-    for (int c = 0; c < nbits; ++c) {
-      float v = wk[(int64_t)(cw * 32 + c) * T];
-      if (v < 0.f) bits |= 1u << c;       // inverted convention
-      else ++pop_real;
+    int pop_real = 0;
+    for (int cw = 0; cw < CW; ++cw) {
+      int nbits = min(32, C - cw * 32);
+      uint32_t bits = 0;
+      for (int c = 0; c < nbits; ++c) {
+        float v = wk[(int64_t)(cw * 32 + c) * T];
+        if (v < 0.f) bits |= 1u << c;     // inverted convention
+        else ++pop_real;
+      }
+      // garbage tail bits = 1 (XOR with a-pack's 0 counts exactly 1)
+      if (nbits < 32) bits |= ~((1u << nbits) - 1u);
+      wp[i * CW + cw] = bits;
     }
-    // garbage tail bits = 1 (so XOR with a-pack's 0 counts exactly 1)
-    if (nbits < 32) bits |= ~((nbits >= 32) ? 0xffffffffu : ((1u << nbits) - 1u));
-    wp[i] = bits;
-    // S[k][t] = C - 2 * popc_real(tap): reduce across cw via atomic
-    atomicAdd(&stab[(int64_t)k * T + t], float(nbits - 2 * pop_real));
+    stab[i] = float(C - 2 * pop_real);
   }
 }
 
@@ -200,10 +201,9 @@ extern "C" void bdbnn_weight_pack(const float* w, uint32_t* wp, float* alpha,
                                   int CW, hipStream_t stream) {
   int64_t per_k = (int64_t)C * KH * KW;
  hipLaunchKernelGGL(( weight_alpha_kernel), dim3(K), dim3(256), 0, stream, w, alpha, K, per_k);
-  int64_t n_words = (int64_t)K * KH * KW * CW;
+  int64_t n_taps = (int64_t)K * KH * KW;
   int block = 256;
-  int grid = (int)bd_min<int64_t>((n_words + block - 1) / block, 65535 * 8);
-  hipMemsetAsync(stab, 0, sizeof(float) * K * KH * KW, stream);
+  int grid = (int)bd_min<int64_t>((n_taps + block - 1) / block, 4096);
  hipLaunchKernelGGL(( weight_pack_kernel), dim3(grid), dim3(block), 0, stream, w, wp, stab, K, C, KH, KW, CW);
 }
 

@@ -159,6 +159,16 @@ class _HardBinaryConvBase(nn.Module):
                 f"kernel_size={self.kernel_size}, stride={self.stride}, "
                 f"padding={self.padding}, act={self.act_mode}")
 
+    def _apply(self, fn, recurse=True):
+        # keep the latent weight NCHW-contiguous even when the model is
+        # converted to channels_last: only our own kernels consume it
+        # (pack expects NCHW) and a channels_last weight would force a
+        # layout copy every step.
+        super()._apply(fn, recurse)
+        if not self.weight.data.is_contiguous():
+            self.weight.data = self.weight.data.contiguous()
+        return self
+
     def forward(self, x):
         t = float(self.t) if self.t is not None else None
         k = float(self.k) if self.k is not None else None
