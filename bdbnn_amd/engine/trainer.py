@@ -20,6 +20,7 @@ import torch
 import torch.nn as nn
 
 from .. import utils
+from ..utils.metrics import MetricsWriter, trace_range
 from ..ops.binary_conv import _HardBinaryConvBase
 from ..ops.kurtosis import KurtosisWeight, kurtosis_loss_fused
 from ..ops.kd import DistributionLoss, WeightKDLoss
@@ -143,6 +144,9 @@ class Trainer:
         self.best_acc1 = 0.0
         self.best_epoch = -1
         self.start_epoch = getattr(args, "start_epoch", 0)
+        self.writer = None
+        if rank == 0 and getattr(args, "log_path", None):
+            self.writer = MetricsWriter(args.log_path)
 
     # ---------------- loss assembly ----------------
 
@@ -218,16 +222,19 @@ class Trainer:
             if self.use_cuda:
                 images = images.contiguous(memory_format=torch.channels_last)
 
-            with torch.autocast("cuda", dtype=torch.bfloat16, enabled=self.amp):
+            with trace_range("fwd+loss"), torch.autocast(
+                    "cuda", dtype=torch.bfloat16, enabled=self.amp):
                 if ts:
                     total, ce, kurt, output = self._step_losses_ts(images, target, epoch)
                 else:
                     total, ce, kurt, output = self._step_losses(images, target, epoch)
 
             self.optimizer.zero_grad(set_to_none=True)
-            total.backward()
-            model.finish_gradient_sync()
-            self.optimizer.step()
+            with trace_range("bwd+allreduce"):
+                total.backward()
+                model.finish_gradient_sync()
+            with trace_range("optimizer"):
+                self.optimizer.step()
 
             acc1, acc5 = utils.accuracy(output, target, topk=(1, 5))
             n = images.size(0)
@@ -244,6 +251,10 @@ class Trainer:
                 progress.display(i)
         for m in (losses, top1, top5):
             m.all_reduce(self.device)
+        if self.writer is not None:
+            self.writer.add_scalar("Train Loss", losses.avg, epoch)
+            self.writer.add_scalar("Train Acc1", top1.avg, epoch)
+            self.writer.add_scalar("Train Acc5", top5.avg, epoch)
         return top1.avg, losses.avg
 
     @torch.no_grad()
@@ -270,6 +281,10 @@ class Trainer:
         if self.rank == 0:
             log.info(f"Val epoch {epoch}: Acc@1 {top1.avg:.3f} "
                      f"Acc@5 {top5.avg:.3f} Loss {losses.avg:.4e}")
+            if self.writer is not None:
+                self.writer.add_scalar("Val Loss", losses.avg, epoch)
+                self.writer.add_scalar("Val Acc1", top1.avg, epoch)
+                self.writer.add_scalar("Val Acc5", top5.avg, epoch)
         return top1.avg
 
     def fit(self, train_loader, val_loader):
@@ -285,6 +300,9 @@ class Trainer:
             if self.rank == 0:
                 log.info(f"***** Best Acc@1 {self.best_acc1:.3f} "
                          f"(epoch {self.best_epoch})")
+                if self.writer is not None:
+                    self.writer.add_scalar("Best val Acc1", self.best_acc1,
+                                           epoch)
                 save_state(self.model, self.optimizer, epoch, a.arch,
                            self.best_acc1, is_best, a.log_path)
         return self.best_acc1

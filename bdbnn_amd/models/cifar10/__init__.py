@@ -7,6 +7,7 @@ num_classes is a kwarg so the same names serve cifar100.
 """
 
 from ...ops.binary_conv import HardBinaryConv_cifar
+from ..vgg_small import vgg_small
 from ..resnet_common import (
     CifarResNet,
     BiBasicBlock,

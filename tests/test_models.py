@@ -75,3 +75,11 @@ def test_react_variant_uses_rprelu():
     from bdbnn_amd.models.resnet_common import RPReLU
     m = models.imagenet.resnet18_react(False)
     assert any(isinstance(mod, RPReLU) for mod in m.modules())
+
+
+def test_vgg_small():
+    m = models.cifar10.vgg_small()
+    out = m(torch.randn(2, 3, 32, 32))
+    assert out.shape == (2, 10)
+    out.sum().backward()
+    assert len(_binary_convs(m)) == 5

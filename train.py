@@ -106,6 +106,9 @@ def build_parser():
                    help="bf16 autocast for the dense (stem/head/backward) path")
     p.add_argument("--synthetic-data", action="store_true",
                    help="synthetic dataset of the selected shape (offline)")
+    p.add_argument("--auto-resume", action="store_true",
+                   help="resume from <log_path>/checkpoint.pth.tar if present "
+                        "(crash recovery; the reference requires manual --resume)")
     return p
 
 
@@ -220,6 +223,12 @@ def _run(args):
 
     if args.resume and os.path.isfile(args.resume):
         trainer.resume(args.resume, reset_resume=args.reset_resume)
+    elif args.auto_resume:
+        auto_ckpt = os.path.join(os.path.dirname(os.path.dirname(args.log_path)),
+                                 "checkpoint.pth.tar")
+        if os.path.isfile(auto_ckpt):
+            log.info(f"auto-resume from {auto_ckpt}")
+            trainer.resume(auto_ckpt)
     if args.resume_teacher and teacher is not None and \
             os.path.isfile(args.resume_teacher):
         from bdbnn_amd.engine.checkpoint import load_state
