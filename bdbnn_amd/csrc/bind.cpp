@@ -348,7 +348,16 @@ std::vector<at::Tensor> bn_act_fwd_train(
                     C, (float)(n / C), (float)momentum, (float)eps,
                     cur_stream());
   auto out = at::empty_like(xc, xc.options(), at::MemoryFormat::ChannelsLast);
-  auto z = at::empty_like(xc, xc.options(), at::MemoryFormat::ChannelsLast);
+  // PReLU backward needs the true pre-activation z; ReLU only needs its
+  // sign (recoverable from out) and identity ignores it — skip the write.
+  at::Tensor z;
+  void* zptr = nullptr;
+  if (act_kind == 1) {
+    z = at::empty_like(xc, xc.options(), at::MemoryFormat::ChannelsLast);
+    zptr = z.data_ptr();
+  } else {
+    z = out;
+  }
   at::Tensor skipc;
   const void* skip_ptr = nullptr;
   if (skip.has_value()) {
@@ -364,7 +373,7 @@ std::vector<at::Tensor> bn_act_fwd_train(
                        a_ptr = af.data_ptr<float>(); }
   bdbnn_bn_act_fwd(xc.data_ptr(), skip_ptr, mean.data_ptr<float>(),
                    invstd.data_ptr<float>(), gf.data_ptr<float>(),
-                   bf.data_ptr<float>(), a_ptr, out.data_ptr(), z.data_ptr(),
+                   bf.data_ptr<float>(), a_ptr, out.data_ptr(), zptr,
                    n, C, (int)act_kind, bf16, cur_stream());
   return {out, z, mean, invstd};
 }

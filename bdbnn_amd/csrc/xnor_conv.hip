@@ -19,16 +19,22 @@
 //   dot(sp,k) = 2*POP + BASE - sum_{t invalid(sp)} S[k][t],
 //   BASE = -2*G - C*T,  G = (32*CW - C) * T.
 //
-// Tiling: 256-thread block computes a 128(spatial) x 64(channel) tile;
-// both operands staged in LDS in 8-word chunks; each thread owns an
-// 8(spatial) x 4(channel) register tile read via ds_read_b128 (a-words
-// broadcast within a 16-lane group, b4 conflict-free across the 256-B
-// bank row): 1024 binary MACs per ~70 instructions.
+// Structure (guide T14 "issue-early / write-late" + one barrier/chunk):
+// 256-thread block computes a 128(spatial) x 64(channel) tile; per
+// 8-word chunk each thread issues its next-chunk global loads FIRST,
+// computes the current LDS buffers (8x4 register tile, ds_read_b128,
+// 1024 binary MACs per ~70 instructions), then writes the prefetched
+// words into the alternate LDS buffer — HBM latency hides under the
+// popcount work.  Word->(tap, offset) address tables are precomputed
+// per block (no integer div/mod in the staging loop).
 #include "common.h"
 
 #define TILE_M 128
 #define TILE_K 64
 #define CHUNK 8
+#define A_PER_THREAD 4  // CHUNK*TILE_M / 256
+#define W_PER_THREAD 2  // CHUNK*TILE_K / 256
+#define MAX_WORDS 160   // KH*KW*CW <= 9*16 (C<=512, 3x3); guarded on host
 
 struct XnorConvParams {
   int N, H, W, C, K, KH, KW, stride, pad, Ho, Wo, CW;
@@ -59,21 +65,23 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
   const int64_t m0 = (int64_t)m_blk * TILE_M;
   const int k0_blk = k_blk * TILE_K;
 
-  __shared__ uint32_t a_lds[CHUNK][TILE_M];
-  __shared__ uint32_t w_lds[CHUNK][TILE_K];
-  __shared__ int row_base[TILE_M];    // input pixel index of tap (0,0)
+  __shared__ uint32_t a_lds[2][CHUNK][TILE_M];
+  __shared__ uint32_t w_lds[2][CHUNK][TILE_K];
+  __shared__ int row_basecw[TILE_M];  // pixel index of tap (0,0), x CW
   __shared__ unsigned short row_inv[TILE_M];  // invalid-tap bitmask (T<=9)
+  __shared__ int off_tab[MAX_WORDS];          // (kh*W+kw)*CW + cw per word
+  __shared__ unsigned char tap_tab[MAX_WORDS];
 
-  // ---- per-row metadata (once per block) ----
+  // ---- per-block metadata ----
   for (int r = tid; r < TILE_M; r += blockDim.x) {
     int64_t sp = m0 + r;
-    if (sp >= M) { row_base[r] = 0; row_inv[r] = 0xffff; continue; }
+    if (sp >= M) { row_basecw[r] = 0; row_inv[r] = 0xffff; continue; }
     int n = int(sp / ((int64_t)p.Ho * p.Wo));
     int rem = int(sp % ((int64_t)p.Ho * p.Wo));
     int oy = rem / p.Wo, ox = rem % p.Wo;
     int iy0 = oy * p.stride - p.pad;
     int ix0 = ox * p.stride - p.pad;
-    row_base[r] = (n * p.H + iy0) * p.W + ix0;
+    row_basecw[r] = ((n * p.H + iy0) * p.W + ix0) * p.CW;
     unsigned short inv = 0;
     for (int t = 0; t < p.T; ++t) {
       int kh = t / p.KW, kw = t % p.KW;
@@ -82,6 +90,12 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
     }
     row_inv[r] = inv;
   }
+  for (int wdx = tid; wdx < p.WORDS; wdx += blockDim.x) {
+    int t = wdx / p.CW, cw = wdx - t * p.CW;
+    int kh = t / p.KW, kw = t - kh * p.KW;
+    off_tab[wdx] = (kh * p.W + kw) * p.CW + cw;
+    tap_tab[wdx] = (unsigned char)t;
+  }
   __syncthreads();
 
   // thread's register tile: rows r0..r0+7, channels kq..kq+3
@@ -89,53 +103,82 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
   const int kq = (tid % 16) * 4;
   int acc[8][4] = {};
 
+  // staging coordinates (fixed per thread)
+  int a_c[A_PER_THREAD], a_r[A_PER_THREAD];
+#pragma unroll
+  for (int it = 0; it < A_PER_THREAD; ++it) {
+    int j = tid + it * 256;
+    a_c[it] = j / TILE_M;
+    a_r[it] = j % TILE_M;
+  }
+  int w_c[W_PER_THREAD], w_k[W_PER_THREAD];
+#pragma unroll
+  for (int it = 0; it < W_PER_THREAD; ++it) {
+    int j = tid + it * 256;
+    w_c[it] = j / TILE_K;
+    w_k[it] = j % TILE_K;
+  }
+
   const int n_chunks = (p.WORDS + CHUNK - 1) / CHUNK;
+  uint32_t av[A_PER_THREAD], wv[W_PER_THREAD];
+
+#define STAGE_LOAD(w0)                                                    \
+  {                                                                       \
+    _Pragma("unroll")                                                     \
+    for (int it = 0; it < A_PER_THREAD; ++it) {                           \
+      int widx = (w0) + a_c[it];                                          \
+      int r = a_r[it];                                                    \
+      uint32_t v = 0;                                                     \
+      if (widx < p.WORDS && !((row_inv[r] >> tap_tab[widx]) & 1))         \
+        v = xp[(int64_t)row_basecw[r] + off_tab[widx]];                   \
+      av[it] = v;                                                         \
+    }                                                                     \
+    _Pragma("unroll")                                                     \
+    for (int it = 0; it < W_PER_THREAD; ++it) {                           \
+      int widx = (w0) + w_c[it];                                          \
+      int kg = k0_blk + w_k[it];                                          \
+      uint32_t v = 0;                                                     \
+      if (widx < p.WORDS && kg < p.K)                                     \
+        v = wp[(int64_t)kg * p.WORDS + widx];                             \
+      wv[it] = v;                                                         \
+    }                                                                     \
+  }
+
+#define STAGE_WRITE(buf)                                                  \
+  {                                                                       \
+    _Pragma("unroll")                                                     \
+    for (int it = 0; it < A_PER_THREAD; ++it)                             \
+      a_lds[buf][a_c[it]][a_r[it]] = av[it];                              \
+    _Pragma("unroll")                                                     \
+    for (int it = 0; it < W_PER_THREAD; ++it)                             \
+      w_lds[buf][w_c[it]][w_k[it]] = wv[it];                              \
+  }
+
+  STAGE_LOAD(0);
+  STAGE_WRITE(0);
+  __syncthreads();
+
   for (int ch = 0; ch < n_chunks; ++ch) {
-    const int w0 = ch * CHUNK;
-    // ---- stage: 4 a-words + 2 w-words per thread ----
-#pragma unroll
-    for (int it = 0; it < 4; ++it) {
-      int j = tid + it * 256;
-      int c = j / TILE_M, r = j % TILE_M;
-      int widx = w0 + c;
-      uint32_t av = 0;
-      if (widx < p.WORDS) {
-        int t = widx / p.CW, cw = widx - t * p.CW;
-        if (!((row_inv[r] >> t) & 1)) {
-          int kh = t / p.KW, kw = t - kh * p.KW;
-          int64_t pix = (int64_t)row_base[r] + kh * p.W + kw;
-          av = xp[pix * p.CW + cw];
-        }
-      }
-      a_lds[c][r] = av;
-    }
-#pragma unroll
-    for (int it = 0; it < 2; ++it) {
-      int j = tid + it * 256;
-      int c = j / TILE_K, kk = j % TILE_K;
-      int widx = w0 + c;
-      int kg = k0_blk + kk;
-      uint32_t wv = 0;
-      if (widx < p.WORDS && kg < p.K)
-        wv = wp[(int64_t)kg * p.WORDS + widx];
-      w_lds[c][kk] = wv;
-    }
-    __syncthreads();
-    // ---- compute ----
+    const int buf = ch & 1;
+    const bool more = ch + 1 < n_chunks;
+    if (more) STAGE_LOAD((ch + 1) * CHUNK);          // issue early
 #pragma unroll
     for (int c = 0; c < CHUNK; ++c) {
       uint32_t a8[8], b4[4];
-      *(uint4*)&a8[0] = *(const uint4*)&a_lds[c][r0];
-      *(uint4*)&a8[4] = *(const uint4*)&a_lds[c][r0 + 4];
-      *(uint4*)b4 = *(const uint4*)&w_lds[c][kq];
+      *(uint4*)&a8[0] = *(const uint4*)&a_lds[buf][c][r0];
+      *(uint4*)&a8[4] = *(const uint4*)&a_lds[buf][c][r0 + 4];
+      *(uint4*)b4 = *(const uint4*)&w_lds[buf][c][kq];
 #pragma unroll
       for (int i = 0; i < 8; ++i)
 #pragma unroll
         for (int j = 0; j < 4; ++j)
           acc[i][j] += __popc(a8[i] ^ b4[j]);
     }
+    if (more) STAGE_WRITE(buf ^ 1);                  // write late
     __syncthreads();
   }
+#undef STAGE_LOAD
+#undef STAGE_WRITE
 
   // ---- epilogue: scale, pad-correction, store ----
   float al[4];
@@ -197,6 +240,7 @@ extern "C" void bdbnn_xnor_conv_fwd(
   p.CW = (C + 31) / 32;
   p.T = KH * KW;
   p.WORDS = p.T * p.CW;
+  if (p.WORDS > MAX_WORDS) abort();  // python guards shapes (<= 3x3, C<=512)
   int G = (32 * p.CW - C) * p.T;
   p.base = -2 * G - C * p.T;
   int64_t M = (int64_t)N * Ho * Wo;
