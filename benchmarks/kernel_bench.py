@@ -6,9 +6,13 @@ under profiles/.
   python benchmarks/kernel_bench.py
 """
 
+import os
+import sys
 import time
 
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from bdbnn_amd import _C
 from bdbnn_amd.ops.binarize import binsign, weight_scale
