@@ -36,6 +36,11 @@ void bdbnn_mask_mul_packed(const void*, const uint32_t*, void*, int64_t,
                            int, int, bool, bool, hipStream_t);
 void bdbnn_weight_decode(const uint32_t*, const float*, void*, int, int,
                          int, int, bool, hipStream_t);
+void bdbnn_maxpool_fwd(const void*, void*, unsigned char*, int, int, int,
+                       int, int, int, int, int, int, bool, hipStream_t);
+void bdbnn_maxpool_bwd(const void*, const unsigned char*, void*, int, int,
+                       int, int, int, int, int, int, int, bool,
+                       hipStream_t);
 void bdbnn_prelu_fwd(const void*, const float*, void*, int64_t, int, bool,
                      hipStream_t);
 void bdbnn_prelu_bwd(const void*, const void*, const float*, void*, float*,
@@ -285,6 +290,40 @@ at::Tensor xnor_conv_fwd(const at::Tensor& xp, const at::Tensor& wp,
                       out.data_ptr(), out_bf16, N, H, W, (int)C, K, KH, KW,
                       (int)stride, (int)pad, Ho, Wo, cur_stream());
   return out;
+}
+
+// ---------------- maxpool ----------------
+
+std::vector<at::Tensor> maxpool_fwd(const at::Tensor& x, int64_t ks,
+                                    int64_t stride, int64_t pad) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4, "maxpool_fwd: 4-D CUDA tensor");
+  auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
+  int N = (int)x.size(0), C = (int)x.size(1);
+  int H = (int)x.size(2), W = (int)x.size(3);
+  int Ho = (int)((H + 2 * pad - ks) / stride + 1);
+  int Wo = (int)((W + 2 * pad - ks) / stride + 1);
+  auto out = at::empty({N, C, Ho, Wo}, xc.options(),
+                       at::MemoryFormat::ChannelsLast);
+  auto idx = at::empty({N, Ho, Wo, C}, xc.options().dtype(at::kByte));
+  bdbnn_maxpool_fwd(xc.data_ptr(), out.data_ptr(),
+                    idx.data_ptr<unsigned char>(), N, C, H, W, Ho, Wo,
+                    (int)ks, (int)stride, (int)pad, is_bf16(xc),
+                    cur_stream());
+  return {out, idx};
+}
+
+at::Tensor maxpool_bwd(const at::Tensor& dy, const at::Tensor& idx,
+                       int64_t H, int64_t W, int64_t ks, int64_t stride,
+                       int64_t pad) {
+  auto dyc = dy.contiguous(at::MemoryFormat::ChannelsLast);
+  int N = (int)dy.size(0), C = (int)dy.size(1);
+  int Ho = (int)dy.size(2), Wo = (int)dy.size(3);
+  auto dx = at::empty({N, C, H, W}, dyc.options(),
+                      at::MemoryFormat::ChannelsLast);
+  bdbnn_maxpool_bwd(dyc.data_ptr(), idx.data_ptr<unsigned char>(),
+                    dx.data_ptr(), N, C, (int)H, (int)W, Ho, Wo, (int)ks,
+                    (int)stride, (int)pad, is_bf16(dyc), cur_stream());
+  return dx;
 }
 
 // ---------------- prelu ----------------
@@ -574,6 +613,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mask_mul_packed", &mask_mul_packed, "dx = mask_bit ? g : 0");
   m.def("weight_decode", &weight_decode, "packed weights -> alpha*(+-1)");
   m.def("xnor_conv_fwd", &xnor_conv_fwd, "bit-packed XNOR+popcount conv");
+  m.def("maxpool_fwd", &maxpool_fwd, "fused NHWC maxpool fwd (+u8 idx)");
+  m.def("maxpool_bwd", &maxpool_bwd, "gather-based NHWC maxpool bwd");
   m.def("prelu_fwd", &prelu_fwd, "fused NHWC per-channel PReLU fwd");
   m.def("prelu_bwd", &prelu_bwd, "fused NHWC per-channel PReLU bwd");
   m.def("bn_act_fwd_train", &bn_act_fwd_train,

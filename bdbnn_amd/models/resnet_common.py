@@ -28,6 +28,7 @@ from ..ops.binary_conv import (
 from ..ops.binarize import LearnableBias
 from ..ops.activations import ChannelPReLU
 from ..ops.bn_act import fused_bn_act
+from ..ops.pool import FusedMaxPool2d
 
 
 class FusedDownsample(nn.Sequential):
@@ -124,7 +125,7 @@ class ResNet(nn.Module):
             self.conv1 = nn.Conv2d(3, width, 3, 1, 1, bias=False)
         self.bn1 = nn.BatchNorm2d(width)
         self.relu = nn.ReLU(inplace=True)
-        self.maxpool = (nn.MaxPool2d(3, 2, 1) if stem == "imagenet"
+        self.maxpool = (FusedMaxPool2d(3, 2, 1) if stem == "imagenet"
                         else nn.Identity())
         stages = [width * (2 ** i) for i in range(len(layers))]
         strides = [1] + [2] * (len(layers) - 1)

@@ -23,6 +23,7 @@ sources = [os.path.join(CSRC, f) for f in (
     "pack.hip",
     "prelu.hip",
     "bn_act.hip",
+    "pool.hip",
     "xnor_conv.hip",
     "kurtosis.hip",
     "kd.hip",

