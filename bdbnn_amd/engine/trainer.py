@@ -170,11 +170,25 @@ class Trainer:
     def _step_losses_ts(self, images, target, epoch):
         """Teacher-student composition (ref:train.py:602-636):
         total = beta*KD_weight + alpha*KD_logit + w_lambda_ce*CE + kurt;
-        --react zeroes beta and CE (ref:train.py:605-609)."""
+        --react zeroes beta and CE (ref:train.py:605-609).
+
+        On GPU the frozen fp32 teacher's forward runs on a side HIP
+        stream, overlapped with the student's forward (both co-resident
+        in HBM — 288 GB makes that trivial)."""
         a = self.args
-        output = self.model(images)
-        with torch.no_grad():
-            t_out = self.teacher.module(images)
+        if self.use_cuda:
+            if not hasattr(self, "_t_stream"):
+                self._t_stream = torch.cuda.Stream()
+            self._t_stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(self._t_stream), torch.no_grad():
+                t_out = self.teacher.module(images)
+            output = self.model(images)
+            torch.cuda.current_stream().wait_stream(self._t_stream)
+            t_out.record_stream(torch.cuda.current_stream())
+        else:
+            output = self.model(images)
+            with torch.no_grad():
+                t_out = self.teacher.module(images)
         alpha = a.alpha
         beta = 0.0 if a.react else a.beta
         w_ce = 0.0 if a.react else getattr(a, "w_lambda_ce", 1.0)
