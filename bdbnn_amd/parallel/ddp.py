@@ -11,11 +11,11 @@ Our own replacement for the reference's implicit DDP all-reduce
   ``async_op`` immediately, overlapping communication with the rest of
   backward;
 * an 8-GPU MI355X node is a fully connected xGMI clique (7 links x
-  ~153 GB/s per GPU) — ring collectives are per-link bound, so we use
-  few LARGE buckets (default 50 MB; a binary ResNet-18's ~11 M fp32
-  grads form one bucket) rather than many small ones: per-step latency
-  is dominated by launch/latency cost, not link bandwidth, at this
-  model size;
+  ~153 GB/s per GPU) — ring collectives are per-link bound.  Default
+  bucket is 12 MB: a binary ResNet-18's ~11 M fp32 grads form ~4
+  buckets, so the deep layers' all-reduce is in flight while the rest
+  of backward still runs, and each bucket is still large enough to be
+  bandwidth- rather than latency-bound on xGMI;
 * SUM + divide by world size (exact reference semantics: DDP averages).
 
 ``state_dict`` keys carry the ``module.`` prefix — the reference's
@@ -47,7 +47,7 @@ class _Bucket:
 
 
 class BucketedDataParallel(nn.Module):
-    def __init__(self, module: nn.Module, bucket_bytes: int = 50 << 20,
+    def __init__(self, module: nn.Module, bucket_bytes: int = 12 << 20,
                  process_group=None, broadcast_params: bool = True):
         super().__init__()
         self.module = module

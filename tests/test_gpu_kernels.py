@@ -81,6 +81,7 @@ def test_ste_mask_modes():
     (2, 48, 10, 10, 64, 3, 2, 1),     # tail channel word + stride 2
     (1, 512, 7, 7, 512, 3, 1, 1),     # deepest layer shape
     (3, 96, 6, 6, 16, 3, 1, 1),       # small K
+    (2, 16, 8, 8, 16, 3, 1, 1),       # CIFAR stage-1 (C=16, tail word)
 ])
 def test_xnor_conv_matches_fp32_conv(shape):
     N, C, H, W, K, ks, stride, pad = shape
