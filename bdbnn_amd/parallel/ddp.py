@@ -27,7 +27,6 @@ import contextlib
 import torch
 import torch.distributed as dist
 import torch.nn as nn
-from torch._utils import _flatten_dense_tensors, _unflatten_dense_tensors
 
 
 class _Bucket:
