@@ -26,8 +26,9 @@ void bdbnn_ste_mask_mul(const void*, const void*, void*, int64_t, bool, bool,
 void bdbnn_weight_pack(const float*, uint32_t*, float*, float*, int, int,
                        int, int, int, hipStream_t);
 void bdbnn_xnor_conv_fwd(const uint32_t*, const uint32_t*, const float*,
-                         const float*, void*, bool, int, int, int, int, int,
-                         int, int, int, int, int, int, hipStream_t);
+                         const float*, void*, float*, float*, bool, int,
+                         int, int, int, int, int, int, int, int, int, int,
+                         hipStream_t);
 void bdbnn_sign_mask_pack(const void*, uint32_t*, uint32_t*, int64_t, int,
                           int, bool, hipStream_t);
 void bdbnn_decode_packed(const uint32_t*, void*, int64_t, int, int, bool,
@@ -269,10 +270,10 @@ at::Tensor weight_decode(const at::Tensor& wp, const at::Tensor& alpha,
 
 // ---------------- xnor conv ----------------
 
-at::Tensor xnor_conv_fwd(const at::Tensor& xp, const at::Tensor& wp,
-                         const at::Tensor& alpha, const at::Tensor& stab,
-                         int64_t C, int64_t stride, int64_t pad,
-                         bool out_bf16) {
+std::vector<at::Tensor> xnor_conv_fwd(
+    const at::Tensor& xp, const at::Tensor& wp, const at::Tensor& alpha,
+    const at::Tensor& stab, int64_t C, int64_t stride, int64_t pad,
+    bool out_bf16, bool want_stats) {
   TORCH_CHECK(xp.is_cuda() && xp.dim() == 4 && xp.scalar_type() == at::kInt,
               "xnor_conv: packed activations int32 [N,H,W,CW]");
   TORCH_CHECK(wp.dim() == 4, "xnor_conv: packed weights [K,KH,KW,CW]");
@@ -284,12 +285,23 @@ at::Tensor xnor_conv_fwd(const at::Tensor& xp, const at::Tensor& wp,
                        xp.options().dtype(out_bf16 ? at::kBFloat16
                                                    : at::kFloat),
                        at::MemoryFormat::ChannelsLast);
+  at::Tensor s1, s2;
+  float *s1p = nullptr, *s2p = nullptr;
+  if (want_stats) {
+    auto fopt = xp.options().dtype(at::kFloat);
+    s1 = at::empty({K}, fopt);
+    s2 = at::empty({K}, fopt);
+    s1p = s1.data_ptr<float>();
+    s2p = s2.data_ptr<float>();
+  }
   bdbnn_xnor_conv_fwd((const uint32_t*)xp.data_ptr<int>(),
                       (const uint32_t*)wp.data_ptr<int>(),
                       alpha.data_ptr<float>(), stab.data_ptr<float>(),
-                      out.data_ptr(), out_bf16, N, H, W, (int)C, K, KH, KW,
-                      (int)stride, (int)pad, Ho, Wo, cur_stream());
-  return out;
+                      out.data_ptr(), s1p, s2p, out_bf16, N, H, W, (int)C,
+                      K, KH, KW, (int)stride, (int)pad, Ho, Wo,
+                      cur_stream());
+  if (want_stats) return {out, s1, s2};
+  return {out};
 }
 
 // ---------------- maxpool ----------------
@@ -365,19 +377,27 @@ std::vector<at::Tensor> bn_act_fwd_train(
     const c10::optional<at::Tensor>& a,
     c10::optional<at::Tensor> running_mean,
     c10::optional<at::Tensor> running_var, double momentum, double eps,
-    int64_t act_kind) {
+    int64_t act_kind, const c10::optional<at::Tensor>& pre_s1,
+    const c10::optional<at::Tensor>& pre_s2) {
   auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
   int C = (int)x.size(1);
   TORCH_CHECK(C <= 1024, "fused bn: C <= 1024");
   int64_t n = xc.numel();
   bool bf16 = is_bf16(xc);
   auto fopt = xc.options().dtype(at::kFloat);
-  auto s1 = at::empty({C}, fopt);
-  auto s2 = at::empty({C}, fopt);
+  at::Tensor s1, s2;
   auto mean = at::empty({C}, fopt);
   auto invstd = at::empty({C}, fopt);
-  bdbnn_bn_stats(xc.data_ptr(), s1.data_ptr<float>(), s2.data_ptr<float>(),
-                 n, C, bf16, cur_stream());
+  if (pre_s1.has_value()) {
+    // stats already accumulated by the producing conv's epilogue
+    s1 = *pre_s1;
+    s2 = *pre_s2;
+  } else {
+    s1 = at::empty({C}, fopt);
+    s2 = at::empty({C}, fopt);
+    bdbnn_bn_stats(xc.data_ptr(), s1.data_ptr<float>(),
+                   s2.data_ptr<float>(), n, C, bf16, cur_stream());
+  }
   float* rm = running_mean.has_value()
                   ? running_mean->data_ptr<float>() : nullptr;
   float* rv = running_var.has_value()

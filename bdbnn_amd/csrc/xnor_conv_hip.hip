@@ -44,11 +44,15 @@ struct XnorConvParams {
   int base;    // -2*G - C*T
 };
 
+// s1/s2 (nullable): per-out-channel sum / sum-of-squares of the STORED
+// (dtype-rounded) outputs, accumulated in the epilogue — feeds the fused
+// BN directly so BN never re-reads the conv output for its stats pass.
 template <typename TO>
 __global__ __launch_bounds__(256) void xnor_conv_kernel(
     const uint32_t* __restrict__ xp, const uint32_t* __restrict__ wp,
     const float* __restrict__ alpha, const float* __restrict__ stab,
-    TO* __restrict__ out, XnorConvParams p, int grid_m) {
+    TO* __restrict__ out, float* __restrict__ s1, float* __restrict__ s2,
+    XnorConvParams p, int grid_m) {
   // XCD-aware block remap (8 XCDs, private L2s): give each XCD a
   // contiguous run of spatial tiles so neighbouring tiles (sharing input
   // rows) land on one L2.  bijective for any grid size.
@@ -72,6 +76,7 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
   __shared__ unsigned short row_inv[TILE_M];  // invalid-tap bitmask (T<=9)
   __shared__ int off_tab[MAX_WORDS];          // (kh*W+kw)*CW + cw per word
   __shared__ unsigned char tap_tab[MAX_WORDS];
+  __shared__ float csum[2][TILE_K];           // per-channel stats partials
 
   // ---- per-block metadata ----
   for (int r = tid; r < TILE_M; r += blockDim.x) {
@@ -97,6 +102,10 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
     off_tab[wdx] = (kh * p.W + kw) * p.CW + cw;
     tap_tab[wdx] = (unsigned char)t;
   }
+  if (s1 != nullptr)
+    for (int c = tid; c < TILE_K; c += blockDim.x) {
+      csum[0][c] = 0.f; csum[1][c] = 0.f;
+    }
   __syncthreads();
 
   // thread's register tile: rows r0..r0+7, channels kq..kq+3
@@ -188,6 +197,7 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
     int kg = k0_blk + kq + j;
     al[j] = (kg < p.K) ? alpha[kg] : 0.f;
   }
+  float st1[4] = {}, st2[4] = {};
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
     int64_t sp = m0 + r0 + i;
@@ -215,6 +225,12 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
       else
         for (int j = 0; j < 4 && kg + j < p.K; ++j)
           out[sp * p.K + kg + j] = vals[j];
+      if (s1 != nullptr)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float v = bf16_to_f32(vals[j]);   // stats of the ROUNDED value
+          st1[j] += v; st2[j] += v * v;
+        }
     } else {
       float vals[4];
 #pragma unroll
@@ -226,13 +242,34 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
       else
         for (int j = 0; j < 4 && kg + j < p.K; ++j)
           ((float*)out)[sp * p.K + kg + j] = vals[j];
+      if (s1 != nullptr)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          st1[j] += vals[j]; st2[j] += vals[j] * vals[j];
+        }
+    }
+  }
+  if (s1 != nullptr) {
+    __syncthreads();  // csum init visible; stores above done per thread
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      atomicAdd(&csum[0][kq + j], st1[j]);
+      atomicAdd(&csum[1][kq + j], st2[j]);
+    }
+    __syncthreads();
+    for (int c = tid; c < TILE_K; c += blockDim.x) {
+      int kg = k0_blk + c;
+      if (kg < p.K) {
+        if (csum[0][c] != 0.f) atomicAdd(&s1[kg], csum[0][c]);
+        if (csum[1][c] != 0.f) atomicAdd(&s2[kg], csum[1][c]);
+      }
     }
   }
 }
 
 extern "C" void bdbnn_xnor_conv_fwd(
     const uint32_t* xp, const uint32_t* wp, const float* alpha,
-    const float* stab, void* out, bool out_bf16,
+    const float* stab, void* out, float* s1, float* s2, bool out_bf16,
     int N, int H, int W, int C, int K, int KH, int KW, int stride, int pad,
     int Ho, int Wo, hipStream_t stream) {
   XnorConvParams p;
@@ -248,10 +285,14 @@ extern "C" void bdbnn_xnor_conv_fwd(
   int grid_m = int((M + TILE_M - 1) / TILE_M);
   int grid_k = (K + TILE_K - 1) / TILE_K;
   dim3 grid(grid_m * grid_k);
+  if (s1 != nullptr) {
+    hipMemsetAsync(s1, 0, sizeof(float) * K, stream);
+    hipMemsetAsync(s2, 0, sizeof(float) * K, stream);
+  }
   if (out_bf16)
    hipLaunchKernelGGL(( xnor_conv_kernel<uint16_t>), dim3(grid), dim3(256), 0, stream, 
-        xp, wp, alpha, stab, (uint16_t*)out, p, grid_m);
+        xp, wp, alpha, stab, (uint16_t*)out, s1, s2, p, grid_m);
   else
    hipLaunchKernelGGL(( xnor_conv_kernel<float>), dim3(grid), dim3(256), 0, stream, 
-        xp, wp, alpha, stab, (float*)out, p, grid_m);
+        xp, wp, alpha, stab, (float*)out, s1, s2, p, grid_m);
 }

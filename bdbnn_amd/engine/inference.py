@@ -32,7 +32,7 @@ class _PackedConvForward:
         xp = nat.sign_pack_nhwc(xc)
         return nat.xnor_conv_fwd(xp, self.wp, self.alpha, self.stab,
                                  self.C, self.stride, self.padding,
-                                 x.dtype == torch.bfloat16)
+                                 x.dtype == torch.bfloat16, False)[0]
 
 
 class PackedInference:

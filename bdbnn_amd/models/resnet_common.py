@@ -35,6 +35,9 @@ class FusedDownsample(nn.Sequential):
     """conv + BN downsample with the fused BN kernel (names '0'/'1' kept)."""
 
     def forward(self, x):
+        if hasattr(self[0], "forward_with_stats"):
+            out, stats = self[0].forward_with_stats(x)
+            return fused_bn_act(out, self[1], stats=stats)
         return fused_bn_act(self[0](x), self[1])
 
 
@@ -106,8 +109,10 @@ class BiBasicBlock(nn.Module):
 
     def forward(self, x):
         identity = self.downsample(x) if self.downsample is not None else x
-        out = fused_bn_act(self.conv1(x), self.bn1, self.act1, skip=identity)
-        out = fused_bn_act(self.conv2(out), self.bn2, self.act2, skip=out)
+        o1, st1 = self.conv1.forward_with_stats(x)
+        out = fused_bn_act(o1, self.bn1, self.act1, skip=identity, stats=st1)
+        o2, st2 = self.conv2.forward_with_stats(out)
+        out = fused_bn_act(o2, self.bn2, self.act2, skip=out, stats=st2)
         return out
 
 

@@ -20,7 +20,8 @@ class _BinConvBlock(nn.Module):
         self.pool = nn.MaxPool2d(2) if pool else nn.Identity()
 
     def forward(self, x):
-        return self.pool(fused_bn_act(self.conv(x), self.bn, self.act))
+        out, stats = self.conv.forward_with_stats(x)
+        return self.pool(fused_bn_act(out, self.bn, self.act, stats=stats))
 
 
 class VGGSmall(nn.Module):

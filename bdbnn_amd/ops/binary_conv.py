@@ -55,7 +55,8 @@ class BinaryConvFunction(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, x, w, stride, padding, act_mode, t, k):
+    def forward(ctx, x, w, stride, padding, act_mode, t, k,
+                want_stats=False):
         ctx.stride = stride
         ctx.padding = padding
         ctx.act_mode = act_mode
@@ -78,16 +79,28 @@ class BinaryConvFunction(torch.autograd.Function):
             else:
                 xp = nat.sign_pack_nhwc(xc)
                 ctx.save_for_backward(x, w)
-            return nat.xnor_conv_fwd(
+            res = nat.xnor_conv_fwd(
                 xp, wp, alpha, stab, x.shape[1], stride, padding,
-                x.dtype == torch.bfloat16)
+                x.dtype == torch.bfloat16, want_stats)
+            if want_stats:
+                # per-channel (sum, sumsq) of the output, accumulated in
+                # the conv epilogue — consumed by the fused BN (its stats
+                # pass is skipped entirely)
+                ctx.mark_non_differentiable(res[1], res[2])
+                return res[0], res[1], res[2]
+            dummy = res[0].new_empty(0)
+            ctx.mark_non_differentiable(dummy)
+            return res[0], dummy, dummy
         ctx.save_for_backward(x, w)
         xb = binsign(x)
         wb = weight_scale(w) * binsign(w)
-        return F.conv2d(xb, wb, None, stride=stride, padding=padding)
+        out = F.conv2d(xb, wb, None, stride=stride, padding=padding)
+        dummy = out.new_empty(0)
+        ctx.mark_non_differentiable(dummy)
+        return out, dummy, dummy
 
     @staticmethod
-    def backward(ctx, g):
+    def backward(ctx, g, _gs1=None, _gs2=None):
         stride, padding = ctx.stride, ctx.padding
         if ctx.packed:
             w, xp, mp, wp, alpha = ctx.saved_tensors
@@ -103,7 +116,7 @@ class BinaryConvFunction(torch.autograd.Function):
             dx = nat.mask_mul_packed(dxb, mp, ctx.in_channels,
                                      ctx.x_dtype == torch.bfloat16)
             dw = nat.ste_mask_mul(dwb.float(), w, 0, 0.0, 0.0)
-            return dx, dw.to(w.dtype), None, None, None, None, None
+            return dx, dw.to(w.dtype), None, None, None, None, None, None
         x, w = ctx.saved_tensors
         if x.is_cuda:
             nat = _C.native_required()
@@ -131,7 +144,7 @@ class BinaryConvFunction(torch.autograd.Function):
         else:
             dx = dxb * _act_grad_mask(x, ctx.act_mode, ctx.t, ctx.k)
             dw = dwb * (w.abs() <= 1).to(w.dtype)
-        return dx, dw.to(w.dtype), None, None, None, None, None
+        return dx, dw.to(w.dtype), None, None, None, None, None, None
 
 
 class _HardBinaryConvBase(nn.Module):
@@ -172,8 +185,21 @@ class _HardBinaryConvBase(nn.Module):
     def forward(self, x):
         t = float(self.t) if self.t is not None else None
         k = float(self.k) if self.k is not None else None
-        return BinaryConvFunction.apply(
-            x, self.weight, self.stride, self.padding, self.act_mode, t, k)
+        out, _, _ = BinaryConvFunction.apply(
+            x, self.weight, self.stride, self.padding, self.act_mode, t, k,
+            False)
+        return out
+
+    def forward_with_stats(self, x):
+        """(out, (s1, s2)|None): per-out-channel sum/sumsq accumulated in
+        the conv epilogue, for the fused BN that consumes the output."""
+        t = float(self.t) if self.t is not None else None
+        k = float(self.k) if self.k is not None else None
+        want = x.is_cuda and self.training
+        out, s1, s2 = BinaryConvFunction.apply(
+            x, self.weight, self.stride, self.padding, self.act_mode, t, k,
+            want)
+        return out, ((s1, s2) if want else None)
 
 
 class HardBinaryConv(_HardBinaryConvBase):

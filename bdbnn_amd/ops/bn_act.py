@@ -24,11 +24,11 @@ _ACT_NONE, _ACT_PRELU, _ACT_RELU = 0, 1, 2
 class _BNActFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, skip, gamma, beta, a, running_mean, running_var,
-                momentum, eps, act_kind, training):
+                momentum, eps, act_kind, training, s1=None, s2=None):
         nat = _C.native_required()
         out, z, mean, invstd = nat.bn_act_fwd_train(
             x, skip, gamma, beta, a, running_mean, running_var,
-            momentum, eps, act_kind)
+            momentum, eps, act_kind, s1, s2)
         ctx.save_for_backward(x, z, mean, invstd, gamma,
                               a if a is not None else torch.empty(0))
         ctx.act_kind = act_kind
@@ -47,11 +47,14 @@ class _BNActFn(torch.autograd.Function):
                 dskip if ctx.has_skip else None,
                 dgamma, dbeta,
                 da if ctx.has_a else None,
-                None, None, None, None, None, None)
+                None, None, None, None, None, None, None, None)
 
 
-def fused_bn_act(x, bn: nn.BatchNorm2d, act=None, skip=None):
-    """BN(x) (+skip) then act.  act: None | ChannelPReLU | 'relu'."""
+def fused_bn_act(x, bn: nn.BatchNorm2d, act=None, skip=None, stats=None):
+    """BN(x) (+skip) then act.  act: None | ChannelPReLU | 'relu'.
+
+    stats: optional (sum, sumsq) per channel of x, pre-accumulated by the
+    producing conv's epilogue — skips BN's own stats read pass."""
     if isinstance(act, ChannelPReLU):
         act_kind, a = _ACT_PRELU, act.weight
     elif act == "relu":
@@ -59,7 +62,7 @@ def fused_bn_act(x, bn: nn.BatchNorm2d, act=None, skip=None):
     elif act is None:
         act_kind, a = _ACT_NONE, None
     else:  # generic module: apply unfused after BN+add
-        out = fused_bn_act(x, bn, None, skip)
+        out = fused_bn_act(x, bn, None, skip, stats=stats)
         return act(out)
 
     C = x.size(1) if x.dim() == 4 else 0
@@ -68,10 +71,11 @@ def fused_bn_act(x, bn: nn.BatchNorm2d, act=None, skip=None):
     if use_fused and bn.training:
         if bn.track_running_stats and bn.num_batches_tracked is not None:
             bn.num_batches_tracked.add_(1)
+        s1, s2 = stats if stats is not None else (None, None)
         return _BNActFn.apply(
             x, skip, bn.weight, bn.bias, a, bn.running_mean, bn.running_var,
             bn.momentum if bn.momentum is not None else 0.1, bn.eps,
-            act_kind, True)
+            act_kind, True, s1, s2)
     if use_fused and not torch.is_grad_enabled() \
             and bn.running_mean is not None:
         nat = _C.native_required()

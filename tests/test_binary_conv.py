@@ -71,7 +71,8 @@ def test_backward_ede_activation():
     conv2 = HardBinaryConv(4, 4, 3, 1, 1)
     with torch.no_grad():
         conv2.weight.copy_(conv.weight)
-    out2 = BinaryConvFunction.apply(x2, conv2.weight, 1, 1, "ste", 2.0, 1.0)
+    out2, _, _ = BinaryConvFunction.apply(x2, conv2.weight, 1, 1, "ste",
+                                          2.0, 1.0)
     out2.sum().backward()
     assert torch.allclose(x.grad, x2.grad, atol=1e-6)
 

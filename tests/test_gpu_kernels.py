@@ -91,7 +91,8 @@ def test_xnor_conv_matches_fp32_conv(shape):
     nat = _nat()
     xp = nat.sign_pack_nhwc(_cl(x))
     wp, alpha, stab = nat.weight_pack(w)
-    out = nat.xnor_conv_fwd(xp, wp, alpha, stab, C, stride, pad, False)
+    out = nat.xnor_conv_fwd(xp, wp, alpha, stab, C, stride, pad, False,
+                            False)[0]
     ref = F.conv2d(binsign(x), weight_scale(w) * binsign(w), None,
                    stride=stride, padding=pad)
     # integer dot + fp32 scale: exact up to fp32 rounding of alpha*int
@@ -106,7 +107,7 @@ def test_xnor_conv_bf16_out():
     nat = _nat()
     xp = nat.sign_pack_nhwc(_cl(x))
     wp, alpha, stab = nat.weight_pack(w)
-    out = nat.xnor_conv_fwd(xp, wp, alpha, stab, 64, 1, 1, True)
+    out = nat.xnor_conv_fwd(xp, wp, alpha, stab, 64, 1, 1, True, False)[0]
     assert out.dtype == torch.bfloat16
     ref = F.conv2d(binsign(x), weight_scale(w) * binsign(w), None, 1, 1)
     assert torch.allclose(out.float(), ref, atol=0.05, rtol=0.02)
@@ -119,13 +120,13 @@ def test_binary_conv_autograd_gpu_vs_cpu():
 
     xg = _cl(x.cuda()).requires_grad_(True)
     wg = w.cuda().requires_grad_(True)
-    out_g = BinaryConvFunction.apply(xg, wg, 1, 1, "ste", None, None)
+    out_g, _, _ = BinaryConvFunction.apply(xg, wg, 1, 1, "ste", None, None)
     g = torch.randn_like(out_g)
     out_g.backward(g)
 
     xc = x.clone().requires_grad_(True)
     wc = w.clone().requires_grad_(True)
-    out_c = BinaryConvFunction.apply(xc, wc, 1, 1, "ste", None, None)
+    out_c, _, _ = BinaryConvFunction.apply(xc, wc, 1, 1, "ste", None, None)
     out_c.backward(g.cpu())
 
     assert torch.allclose(out_g.cpu(), out_c, atol=1e-3, rtol=1e-4)
@@ -431,3 +432,44 @@ def test_fused_maxpool_matches_torch(dtype):
     else:
         assert torch.allclose(x.grad.float().sum(), x2.grad.float().sum(),
                               rtol=1e-2)
+
+
+def test_conv_epilogue_stats_match_output_sums():
+    torch.manual_seed(17)
+    nat = _nat()
+    x = torch.randn(3, 64, 14, 14, device="cuda")
+    w = torch.randn(96, 64, 3, 3, device="cuda")
+    xp = nat.sign_pack_nhwc(_cl(x))
+    wp, alpha, stab = nat.weight_pack(w)
+    out, s1, s2 = nat.xnor_conv_fwd(xp, wp, alpha, stab, 64, 1, 1, False,
+                                    True)
+    ref1 = out.sum(dim=(0, 2, 3))
+    ref2 = (out * out).sum(dim=(0, 2, 3))
+    assert torch.allclose(s1, ref1, rtol=1e-4, atol=1e-2)
+    assert torch.allclose(s2, ref2, rtol=1e-4, atol=1e-1)
+
+
+def test_block_with_fused_stats_matches_composition():
+    """A whole BiBasicBlock forward+backward with conv-epilogue stats vs
+    the CPU oracle."""
+    from bdbnn_amd.models.resnet_common import BiBasicBlock
+    torch.manual_seed(18)
+    blk_cpu = BiBasicBlock(64, 64)
+    blk_gpu = BiBasicBlock(64, 64)
+    blk_gpu.load_state_dict(blk_cpu.state_dict())
+    blk_gpu = blk_gpu.cuda().to(memory_format=torch.channels_last)
+    x = torch.randn(2, 64, 10, 10)
+    xg = _cl(x.cuda()).requires_grad_(True)
+    xc = x.clone().requires_grad_(True)
+    out_g = blk_gpu(xg)
+    out_c = blk_cpu(xc)
+    assert torch.allclose(out_g.cpu(), out_c, atol=5e-3, rtol=1e-3), \
+        (out_g.cpu() - out_c).abs().max().item()
+    g = torch.randn_like(out_c)
+    out_g.backward(_cl(g.cuda()))
+    out_c.backward(g)
+    assert torch.allclose(xg.grad.cpu(), xc.grad, atol=5e-3, rtol=1e-2)
+    for (n1, p1), (n2, p2) in zip(blk_gpu.named_parameters(),
+                                  blk_cpu.named_parameters()):
+        assert torch.allclose(p1.grad.cpu(), p2.grad, atol=5e-3,
+                              rtol=1e-2), n1
