@@ -22,12 +22,16 @@ CPU path: the pure-PyTorch oracle (also the numerics reference for the
 kernels' tests).
 """
 
+import os
+
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
 from .binarize import binsign, binarize_weight, weight_scale, SignSTE, SignEDE, SignApprox
 from .. import _C
+
+_CONV_STATS = os.environ.get("BDBNN_CONV_STATS", "0") == "1"
 
 
 def _act_grad_mask(x: torch.Tensor, mode: str, t, k) -> torch.Tensor:
@@ -192,10 +196,15 @@ class _HardBinaryConvBase(nn.Module):
 
     def forward_with_stats(self, x):
         """(out, (s1, s2)|None): per-out-channel sum/sumsq accumulated in
-        the conv epilogue, for the fused BN that consumes the output."""
+        the conv epilogue, for the fused BN that consumes the output.
+
+        Measured OFF by default: with ~12k blocks all atomically adding
+        into K stat words the contention costs more than the BN stats
+        pass it saves (the conv output is still L2/L3-resident when BN
+        reads it).  BDBNN_CONV_STATS=1 re-enables for experiments."""
         t = float(self.t) if self.t is not None else None
         k = float(self.k) if self.k is not None else None
-        want = x.is_cuda and self.training
+        want = (x.is_cuda and self.training and _CONV_STATS)
         out, s1, s2 = BinaryConvFunction.apply(
             x, self.weight, self.stride, self.padding, self.act_mode, t, k,
             want)
