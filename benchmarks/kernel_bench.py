@@ -50,7 +50,7 @@ def main():
         pad = 1 if ks == 3 else 0
         Ho = (H + 2 * pad - ks) // st + 1
         ms = timeit(lambda: nat.xnor_conv_fwd(xp, wp, alpha, stab, C, st,
-                                              pad, True))
+                                              pad, True, False))
         macs = N * Ho * Ho * K * C * ks * ks
         tmacs = macs / (ms / 1e3) / 1e12
         rows.append((f"{C}x{H}x{H} -> {K}, {ks}x{ks}/s{st}", ms, tmacs))
@@ -85,11 +85,12 @@ def main():
     rm = torch.zeros(C, device="cuda")
     rv = torch.ones(C, device="cuda")
     out = nat.bn_act_fwd_train(xcl, gcl, gamma, beta, a, rm, rv, 0.1,
-                               1e-5, 1)
+                               1e-5, 1, None, None)
     o, z, mean, invstd = out
     row("bn stats+finalize (1R)",
         timeit(lambda: nat.bn_act_fwd_train(xcl, None, gamma, beta, None,
-                                            rm, rv, 0.1, 1e-5, 0)), 3)
+                                            rm, rv, 0.1, 1e-5, 0, None,
+                                            None)), 3)
     row("bn_act_bwd (6R2W)",
         timeit(lambda: nat.bn_act_bwd(gcl, z, xcl, mean, invstd, gamma, a,
                                       1, True)), 8)
