@@ -46,7 +46,7 @@ struct XnorConvParams {
 // s1/s2 (nullable): per-out-channel sum / sum-of-squares of the STORED
 // (dtype-rounded) outputs, accumulated in the epilogue — feeds the fused
 // BN directly so BN never re-reads the conv output for its stats pass.
-template <typename TO>
+template <typename TO, bool STATS>
 __global__ __launch_bounds__(256) void xnor_conv_kernel(
     const uint32_t* __restrict__ xp, const uint32_t* __restrict__ wp,
     const float* __restrict__ alpha, const float* __restrict__ stab,
@@ -101,7 +101,7 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
     off_tab[wdx] = (kh * p.W + kw) * p.CW + cw;
     tap_tab[wdx] = (unsigned char)t;
   }
-  if (s1 != nullptr)
+  if constexpr (STATS)
     for (int c = tid; c < TILE_K; c += blockDim.x) {
       csum[0][c] = 0.f; csum[1][c] = 0.f;
     }
@@ -196,7 +196,7 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
     int kg = k0_blk + kq + j;
     al[j] = (kg < p.K) ? alpha[kg] : 0.f;
   }
-  float st1[4] = {}, st2[4] = {};
+  float st1[STATS ? 4 : 1] = {}, st2[STATS ? 4 : 1] = {};
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
     int64_t sp = m0 + r0 + i;
@@ -224,7 +224,7 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
       else
         for (int j = 0; j < 4 && kg + j < p.K; ++j)
           out[sp * p.K + kg + j] = vals[j];
-      if (s1 != nullptr)
+      if constexpr (STATS)
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
           float v = bf16_to_f32(vals[j]);   // stats of the ROUNDED value
@@ -241,14 +241,14 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
       else
         for (int j = 0; j < 4 && kg + j < p.K; ++j)
           ((float*)out)[sp * p.K + kg + j] = vals[j];
-      if (s1 != nullptr)
+      if constexpr (STATS)
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
           st1[j] += vals[j]; st2[j] += vals[j] * vals[j];
         }
     }
   }
-  if (s1 != nullptr) {
+  if constexpr (STATS) {
     __syncthreads();  // csum init visible; stores above done per thread
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
@@ -288,10 +288,19 @@ extern "C" void bdbnn_xnor_conv_fwd(
     hipMemsetAsync(s1, 0, sizeof(float) * K, stream);
     hipMemsetAsync(s2, 0, sizeof(float) * K, stream);
   }
-  if (out_bf16)
-    xnor_conv_kernel<uint16_t><<<grid, 256, 0, stream>>>(
-        xp, wp, alpha, stab, (uint16_t*)out, s1, s2, p, grid_m);
-  else
-    xnor_conv_kernel<float><<<grid, 256, 0, stream>>>(
-        xp, wp, alpha, stab, (float*)out, s1, s2, p, grid_m);
+  if (s1 != nullptr) {
+    if (out_bf16)
+      xnor_conv_kernel<uint16_t, true><<<grid, 256, 0, stream>>>(
+          xp, wp, alpha, stab, (uint16_t*)out, s1, s2, p, grid_m);
+    else
+      xnor_conv_kernel<float, true><<<grid, 256, 0, stream>>>(
+          xp, wp, alpha, stab, (float*)out, s1, s2, p, grid_m);
+  } else {
+    if (out_bf16)
+      xnor_conv_kernel<uint16_t, false><<<grid, 256, 0, stream>>>(
+          xp, wp, alpha, stab, (uint16_t*)out, s1, s2, p, grid_m);
+    else
+      xnor_conv_kernel<float, false><<<grid, 256, 0, stream>>>(
+          xp, wp, alpha, stab, (float*)out, s1, s2, p, grid_m);
+  }
 }
