@@ -71,7 +71,10 @@ class BinaryConvFunction(torch.autograd.Function):
             nat = _C.native_required()
             xc = x.contiguous(memory_format=torch.channels_last)
             wp, alpha, stab = nat.weight_pack(w)  # bits, alpha[K], S[K][T]
-            if act_mode == "ste" and t is None:
+            if not torch.is_grad_enabled():
+                # inference/validation: no mask plane, nothing saved
+                xp = nat.sign_pack_nhwc(xc)
+            elif act_mode == "ste" and t is None:
                 # packed fast path: sign + clip-STE mask bitplanes in one
                 # pass; the fp activations are NOT saved — backward works
                 # entirely from the 1-bit planes (32x less read traffic)
