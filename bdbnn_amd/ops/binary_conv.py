@@ -71,9 +71,12 @@ class BinaryConvFunction(torch.autograd.Function):
             nat = _C.native_required()
             xc = x.contiguous(memory_format=torch.channels_last)
             wp, alpha, stab = nat.weight_pack(w)  # bits, alpha[K], S[K][T]
-            if not torch.is_grad_enabled():
+            # (grad mode is always off inside Function.forward; use
+            # needs_input_grad to detect inference/no-grad calls)
+            if not (ctx.needs_input_grad[0] or ctx.needs_input_grad[1]):
                 # inference/validation: no mask plane, nothing saved
                 xp = nat.sign_pack_nhwc(xc)
+                ctx.packed = None  # backward must never run
             elif act_mode == "ste" and t is None:
                 # packed fast path: sign + clip-STE mask bitplanes in one
                 # pass; the fp activations are NOT saved — backward works
