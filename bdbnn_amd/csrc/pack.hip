@@ -168,30 +168,36 @@ __global__ void weight_alpha_kernel(const float* __restrict__ w,
 
 __global__ void weight_pack_kernel(const float* __restrict__ w,
                                    uint32_t* __restrict__ wp,
-                                   float* __restrict__ stab,
-                                   int K, int C, int KH, int KW, int CW) {
-  // one thread per (k, tap): packs that tap's CW words AND computes the
-  // pad-correction entry S[k][t] = C - 2*popc_real with no atomics.
-  int T = KH * KW;
+                                   int K, int C, int T, int CW) {
+  // one thread per packed word (k, t, cw); w layout stride of c is T
+  int64_t n_words = (int64_t)K * T * CW;
+  GRID_STRIDE(i, n_words) {
+    int cw = int(i % CW);
+    int t = int((i / CW) % T);
+    int k = int(i / CW / T);
+    int nbits = min(32, C - cw * 32);
+    const float* wk = w + ((int64_t)k * C) * T + t;
+    uint32_t bits = 0;
+    for (int c = 0; c < nbits; ++c)
+      if (wk[(int64_t)(cw * 32 + c) * T] < 0.f)
+        bits |= 1u << c;                  // inverted convention
+    // garbage tail bits = 1 (XOR with a-pack's 0 counts exactly 1)
+    if (nbits < 32) bits |= ~((1u << nbits) - 1u);
+    wp[i] = bits;
+  }
+}
+
+// pad-correction table from the packed (inverted) bits:
+// S[k][t] = C - 2*q_t = 2*popc(inverted real bits) - C.
+__global__ void weight_stab_kernel(const uint32_t* __restrict__ wp,
+                                   float* __restrict__ stab, int K, int C,
+                                   int T, int CW) {
   int64_t n_taps = (int64_t)K * T;
+  int garbage = 32 * CW - C;   // garbage bits are 1 in every tap
   GRID_STRIDE(i, n_taps) {
-    int t = int(i % T);
-    int k = int(i / T);
-    const float* wk = w + ((int64_t)k * C) * T + t;  // + c*T steps
-    int pop_real = 0;
-    for (int cw = 0; cw < CW; ++cw) {
-      int nbits = min(32, C - cw * 32);
-      uint32_t bits = 0;
-      for (int c = 0; c < nbits; ++c) {
-        float v = wk[(int64_t)(cw * 32 + c) * T];
-        if (v < 0.f) bits |= 1u << c;     // inverted convention
-        else ++pop_real;
-      }
-      // garbage tail bits = 1 (XOR with a-pack's 0 counts exactly 1)
-      if (nbits < 32) bits |= ~((1u << nbits) - 1u);
-      wp[i * CW + cw] = bits;
-    }
-    stab[i] = float(C - 2 * pop_real);
+    int pop = 0;
+    for (int cw = 0; cw < CW; ++cw) pop += __popc(wp[i * CW + cw]);
+    stab[i] = float(2 * (pop - garbage) - C);
   }
 }
 
@@ -200,10 +206,14 @@ extern "C" void bdbnn_weight_pack(const float* w, uint32_t* wp, float* alpha,
                                   int CW, hipStream_t stream) {
   int64_t per_k = (int64_t)C * KH * KW;
   weight_alpha_kernel<<<K, 256, 0, stream>>>(w, alpha, K, per_k);
-  int64_t n_taps = (int64_t)K * KH * KW;
+  int T = KH * KW;
+  int64_t n_words = (int64_t)K * T * CW;
   int block = 256;
-  int grid = (int)bd_min<int64_t>((n_taps + block - 1) / block, 4096);
-  weight_pack_kernel<<<grid, block, 0, stream>>>(w, wp, stab, K, C, KH, KW, CW);
+  int grid = (int)bd_min<int64_t>((n_words + block - 1) / block, 4096);
+  weight_pack_kernel<<<grid, block, 0, stream>>>(w, wp, K, C, T, CW);
+  int64_t n_taps = (int64_t)K * T;
+  int grid2 = (int)bd_min<int64_t>((n_taps + block - 1) / block, 4096);
+  weight_stab_kernel<<<grid2, block, 0, stream>>>(wp, stab, K, C, T, CW);
 }
 
 // ---------------- sign+mask pack (one pass) ----------------

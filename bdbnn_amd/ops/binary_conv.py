@@ -125,7 +125,7 @@ class BinaryConvFunction(torch.autograd.Function):
                 1, [True, True, False])[:2]
             dx = nat.mask_mul_packed(dxb, mp, ctx.in_channels,
                                      ctx.x_dtype == torch.bfloat16)
-            dw = nat.ste_mask_mul(dwb.float(), w, 0, 0.0, 0.0)
+            dw = nat.ste_mask_mul(dwb, w, 0, 0.0, 0.0)
             return dx, dw.to(w.dtype), None, None, None, None, None, None
         x, w = ctx.saved_tensors
         if x.is_cuda:
@@ -150,7 +150,7 @@ class BinaryConvFunction(torch.autograd.Function):
             else:
                 t, k = 0.0, 0.0
             dx = nat.ste_mask_mul(dxb, x, mode_id, t, k)
-            dw = nat.ste_mask_mul(dwb.float(), w, 0, 0.0, 0.0)
+            dw = nat.ste_mask_mul(dwb, w, 0, 0.0, 0.0)
         else:
             dx = dxb * _act_grad_mask(x, ctx.act_mode, ctx.t, ctx.k)
             dw = dwb * (w.abs() <= 1).to(w.dtype)
