@@ -42,6 +42,14 @@ void bdbnn_maxpool_fwd(const void*, void*, unsigned char*, int, int, int,
 void bdbnn_maxpool_bwd(const void*, const unsigned char*, void*, int, int,
                        int, int, int, int, int, int, int, bool,
                        hipStream_t);
+void bdbnn_kd_logit_fwd(const void*, const void*, float*, float*, int,
+                        int, bool, hipStream_t);
+void bdbnn_kd_logit_bwd(const void*, const void*, const float*, void*,
+                        float, int, int, bool, hipStream_t);
+void bdbnn_ce_fwd(const void*, const int64_t*, float*, float*, int, int,
+                  bool, hipStream_t);
+void bdbnn_ce_bwd(const void*, const int64_t*, const float*, void*, float,
+                  int, int, bool, hipStream_t);
 void bdbnn_prelu_fwd(const void*, const float*, void*, int64_t, int, bool,
                      hipStream_t);
 void bdbnn_prelu_bwd(const void*, const void*, const float*, void*, float*,
@@ -302,6 +310,61 @@ std::vector<at::Tensor> xnor_conv_fwd(
                       cur_stream());
   if (want_stats) return {out, s1, s2};
   return {out};
+}
+
+// ---------------- fused classification losses ----------------
+
+std::vector<at::Tensor> kd_logit_fwd(const at::Tensor& s,
+                                     const at::Tensor& t) {
+  TORCH_CHECK(s.is_cuda() && s.dim() == 2 && s.sizes() == t.sizes(),
+              "kd_logit: (B,C) logits");
+  auto sc = s.contiguous();
+  auto tc = t.to(s.scalar_type()).contiguous();
+  int B = (int)s.size(0), C = (int)s.size(1);
+  auto fopt = sc.options().dtype(at::kFloat);
+  auto stats = at::empty({B, 4}, fopt);
+  auto out = at::empty({}, fopt);
+  bdbnn_kd_logit_fwd(sc.data_ptr(), tc.data_ptr(), stats.data_ptr<float>(),
+                     out.data_ptr<float>(), B, C, is_bf16(sc),
+                     cur_stream());
+  return {out, stats, tc};
+}
+
+at::Tensor kd_logit_bwd(const at::Tensor& s, const at::Tensor& t,
+                        const at::Tensor& stats, double gscale) {
+  auto sc = s.contiguous();
+  int B = (int)s.size(0), C = (int)s.size(1);
+  auto ds = at::empty_like(sc);
+  bdbnn_kd_logit_bwd(sc.data_ptr(), t.data_ptr(), stats.data_ptr<float>(),
+                     ds.data_ptr(), (float)(gscale / B), B, C, is_bf16(sc),
+                     cur_stream());
+  return ds;
+}
+
+std::vector<at::Tensor> ce_fwd(const at::Tensor& s, const at::Tensor& y) {
+  TORCH_CHECK(s.is_cuda() && s.dim() == 2 && y.scalar_type() == at::kLong,
+              "ce: (B,C) logits + int64 targets");
+  auto sc = s.contiguous();
+  auto yc = y.contiguous();
+  int B = (int)s.size(0), C = (int)s.size(1);
+  auto fopt = sc.options().dtype(at::kFloat);
+  auto stats = at::empty({B}, fopt);
+  auto out = at::empty({}, fopt);
+  bdbnn_ce_fwd(sc.data_ptr(), yc.data_ptr<int64_t>(),
+               stats.data_ptr<float>(), out.data_ptr<float>(), B, C,
+               is_bf16(sc), cur_stream());
+  return {out, stats};
+}
+
+at::Tensor ce_bwd(const at::Tensor& s, const at::Tensor& y,
+                  const at::Tensor& stats, double gscale) {
+  auto sc = s.contiguous();
+  int B = (int)s.size(0), C = (int)s.size(1);
+  auto ds = at::empty_like(sc);
+  bdbnn_ce_bwd(sc.data_ptr(), y.data_ptr<int64_t>(),
+               stats.data_ptr<float>(), ds.data_ptr(),
+               (float)(gscale / B), B, C, is_bf16(sc), cur_stream());
+  return ds;
 }
 
 // ---------------- maxpool ----------------
@@ -633,6 +696,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mask_mul_packed", &mask_mul_packed, "dx = mask_bit ? g : 0");
   m.def("weight_decode", &weight_decode, "packed weights -> alpha*(+-1)");
   m.def("xnor_conv_fwd", &xnor_conv_fwd, "bit-packed XNOR+popcount conv");
+  m.def("kd_logit_fwd", &kd_logit_fwd, "fused logit-KD fwd");
+  m.def("kd_logit_bwd", &kd_logit_bwd, "fused logit-KD bwd");
+  m.def("ce_fwd", &ce_fwd, "fused cross-entropy fwd");
+  m.def("ce_bwd", &ce_bwd, "fused cross-entropy bwd");
   m.def("maxpool_fwd", &maxpool_fwd, "fused NHWC maxpool fwd (+u8 idx)");
   m.def("maxpool_bwd", &maxpool_bwd, "gather-based NHWC maxpool bwd");
   m.def("prelu_fwd", &prelu_fwd, "fused NHWC per-channel PReLU fwd");

@@ -25,6 +25,7 @@ from ..ops.binary_conv import _HardBinaryConvBase
 from ..ops.kurtosis import KurtosisWeight, kurtosis_loss_fused
 from ..ops.kd import DistributionLoss, WeightKDLoss
 from ..ops.optim import build_optimizer
+from ..ops.losses import FusedCrossEntropy
 from ..parallel import BucketedDataParallel
 from .checkpoint import save_state, load_state
 
@@ -134,7 +135,7 @@ class Trainer:
             self.kd_logit = DistributionLoss()
             self.kd_weight = WeightKDLoss(self.model, self.teacher)
 
-        self.criterion = nn.CrossEntropyLoss().to(self.device)
+        self.criterion = FusedCrossEntropy().to(self.device)
         self.optimizer, self.scheduler = build_optimizer(args, self.model)
 
         self.kurt_table = build_kurtosis_table(self.model, args)

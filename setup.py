@@ -24,6 +24,7 @@ sources = [os.path.join(CSRC, f) for f in (
     "prelu.hip",
     "bn_act.hip",
     "pool.hip",
+    "loss.hip",
     "xnor_conv.hip",
     "kurtosis.hip",
     "kd.hip",

@@ -473,3 +473,45 @@ def test_block_with_fused_stats_matches_composition():
                                   blk_cpu.named_parameters()):
         assert torch.allclose(p1.grad.cpu(), p2.grad, atol=5e-3,
                               rtol=1e-2), n1
+
+
+# ---------------- fused classification losses ----------------
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_logit_kd(dtype):
+    from bdbnn_amd.ops.losses import fused_logit_kd
+    torch.manual_seed(19)
+    s = (torch.randn(16, 1000, device="cuda", dtype=dtype) * 3
+         ).requires_grad_(True)
+    t = torch.randn(16, 1000, device="cuda", dtype=dtype) * 3
+    loss = fused_logit_kd(s, t)
+    s2 = s.detach().float().clone().requires_grad_(True)
+    ref = -(torch.softmax(t.float(), 1)
+            * torch.log_softmax(s2, 1)).sum(1).mean()
+    atol = 1e-5 if dtype == torch.float32 else 5e-2
+    assert torch.allclose(loss.float(), ref, atol=atol, rtol=1e-3)
+    loss.backward()
+    ref.backward()
+    assert torch.allclose(s.grad.float(), s2.grad,
+                          atol=1e-5 if dtype == torch.float32 else 1e-3,
+                          rtol=1e-2)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_cross_entropy(dtype):
+    from bdbnn_amd.ops.losses import FusedCrossEntropy
+    torch.manual_seed(20)
+    s = (torch.randn(32, 1000, device="cuda", dtype=dtype) * 2
+         ).requires_grad_(True)
+    y = torch.randint(0, 1000, (32,), device="cuda")
+    loss = FusedCrossEntropy()(s, y)
+    s2 = s.detach().float().clone().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(s2, y)
+    atol = 1e-5 if dtype == torch.float32 else 5e-2
+    assert torch.allclose(loss.float(), ref, atol=atol, rtol=1e-3)
+    g = torch.tensor(1.7, device="cuda")
+    loss.backward(g)
+    ref.backward(g.float())
+    assert torch.allclose(s.grad.float(), s2.grad,
+                          atol=1e-5 if dtype == torch.float32 else 1e-3,
+                          rtol=1e-2)
