@@ -8,7 +8,6 @@ On CPU-only machines (CI) the pure-PyTorch oracle path is used.
 """
 
 import importlib
-import os
 
 _native = None
 _native_err = None

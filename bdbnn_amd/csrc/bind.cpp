@@ -117,7 +117,6 @@ Schedule build_schedule(const std::vector<at::Tensor>& ts,
       bo.push_back(off);
     }
   }
-  auto opts = at::TensorOptions().device(dev);
   Schedule s;
   s.n_blocks = (int)bt.size();
   s.bt = at::from_blob(bt.data(), {(int64_t)bt.size()},

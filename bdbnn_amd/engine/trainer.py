@@ -22,7 +22,7 @@ import torch.nn as nn
 from .. import utils
 from ..utils.metrics import MetricsWriter, trace_range
 from ..ops.binary_conv import _HardBinaryConvBase
-from ..ops.kurtosis import KurtosisWeight, kurtosis_loss_fused
+from ..ops.kurtosis import kurtosis_loss_fused
 from ..ops.kd import DistributionLoss, WeightKDLoss
 from ..ops.optim import build_optimizer
 from ..ops.losses import FusedCrossEntropy

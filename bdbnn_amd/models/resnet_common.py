@@ -20,11 +20,7 @@ matching the 19-entry per-layer kurtosis target lists
 import torch
 import torch.nn as nn
 
-from ..ops.binary_conv import (
-    HardBinaryConv,
-    HardBinaryConv_react,
-    HardBinaryConv_cifar,
-)
+from ..ops.binary_conv import HardBinaryConv
 from ..ops.binarize import LearnableBias
 from ..ops.activations import ChannelPReLU
 from ..ops.bn_act import fused_bn_act

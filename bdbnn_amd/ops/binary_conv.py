@@ -28,7 +28,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from .binarize import binsign, binarize_weight, weight_scale, SignSTE, SignEDE, SignApprox
+from .binarize import binsign, weight_scale
 from .. import _C
 
 _CONV_STATS = os.environ.get("BDBNN_CONV_STATS", "0") == "1"

@@ -18,7 +18,6 @@ z_i/(n-1).
 """
 
 import torch
-import torch.nn as nn
 
 from .. import _C
 
