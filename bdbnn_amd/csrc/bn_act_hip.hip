@@ -134,46 +134,22 @@ __global__ void bn_act_bwd_reduce_kernel(
     av[j] = (ACT == 1) ? a[cc] : 0.f;
   }
   float s0[8] = {}, s1[8] = {}, s2[8] = {};
-  float dyv[2][8], zv[2][8], xv[2][8];
-  // 2-deep pixel unroll: 6 independent 16-B loads in flight per iteration
-  int64_t p = m.p0;
-  for (; p + m.pstep < n_pix; p += 2 * m.pstep) {
-    int64_t i0 = p * C + m.c0;
-    int64_t i1 = (p + m.pstep) * C + m.c0;
-    load8(dy, i0, dyv[0]); load8(dy, i1, dyv[1]);
-    load8(x, i0, xv[0]);   load8(x, i1, xv[1]);
-    if (ACT != 0) { load8(z, i0, zv[0]); load8(z, i1, zv[1]); }
-#pragma unroll
-    for (int u = 0; u < 2; ++u)
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float dz = dyv[u][j];
-        if (ACT == 1) {
-          dz = zv[u][j] > 0.f ? dyv[u][j] : av[j] * dyv[u][j];
-          if (zv[u][j] <= 0.f) s2[j] += dyv[u][j] * zv[u][j];
-        } else if (ACT == 2) {
-          dz = zv[u][j] > 0.f ? dyv[u][j] : 0.f;
-        }
-        float xhat = (xv[u][j] - mu[j]) * is[j];
-        s0[j] += dz;
-        s1[j] += dz * xhat;
-      }
-  }
-  for (; p < n_pix; p += m.pstep) {
+  float dyv[8], zv[8], xv[8];
+  for (int64_t p = m.p0; p < n_pix; p += m.pstep) {
     int64_t i = p * C + m.c0;
-    load8(dy, i, dyv[0]);
-    load8(x, i, xv[0]);
-    if (ACT != 0) load8(z, i, zv[0]);
+    load8(dy, i, dyv);
+    load8(x, i, xv);
+    if (ACT != 0) load8(z, i, zv);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      float dz = dyv[0][j];
+      float dz = dyv[j];
       if (ACT == 1) {
-        dz = zv[0][j] > 0.f ? dyv[0][j] : av[j] * dyv[0][j];
-        if (zv[0][j] <= 0.f) s2[j] += dyv[0][j] * zv[0][j];
+        dz = zv[j] > 0.f ? dyv[j] : av[j] * dyv[j];
+        if (zv[j] <= 0.f) s2[j] += dyv[j] * zv[j];
       } else if (ACT == 2) {
-        dz = zv[0][j] > 0.f ? dyv[0][j] : 0.f;
+        dz = zv[j] > 0.f ? dyv[j] : 0.f;
       }
-      float xhat = (xv[0][j] - mu[j]) * is[j];
+      float xhat = (xv[j] - mu[j]) * is[j];
       s0[j] += dz;
       s1[j] += dz * xhat;
     }
