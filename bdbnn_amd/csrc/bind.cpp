@@ -286,6 +286,9 @@ std::vector<at::Tensor> xnor_conv_fwd(
   TORCH_CHECK(wp.dim() == 4, "xnor_conv: packed weights [K,KH,KW,CW]");
   int N = (int)xp.size(0), H = (int)xp.size(1), W = (int)xp.size(2);
   int K = (int)wp.size(0), KH = (int)wp.size(1), KW = (int)wp.size(2);
+  TORCH_CHECK(KH * KW * wp.size(3) <= 160,
+              "xnor_conv: KH*KW*ceil(C/32) must be <= 160 "
+              "(kernel address-table size; C <= 512 at 3x3)");
   int Ho = (int)((H + 2 * pad - KH) / stride + 1);
   int Wo = (int)((W + 2 * pad - KW) / stride + 1);
   auto out = at::empty({N, K, Ho, Wo},

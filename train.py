@@ -216,8 +216,13 @@ def _run(args):
     model = build_model(args, args.arch, args.custom_resnet)
     teacher = build_teacher(args) if args.imagenet_setting_step_2_ts else None
 
-    device = (torch.device(f"cuda:{local_rank % max(torch.cuda.device_count(),1)}")
-              if torch.cuda.is_available() else torch.device("cpu"))
+    if torch.cuda.is_available():
+        dev_idx = (args.gpu if args.gpu is not None
+                   else local_rank % max(torch.cuda.device_count(), 1))
+        device = torch.device(f"cuda:{dev_idx}")
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
     trainer = Trainer(model, args, teacher=teacher, device=device,
                       world_size=world_size, rank=rank)
 
