@@ -5,6 +5,8 @@
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
+#include <string>
+#include <unordered_map>
 #include <vector>
 
 // ---- mirrors of csrc/common.h (kept in sync) ----
@@ -106,8 +108,8 @@ struct Schedule {
   int n_blocks;
 };
 
-Schedule build_schedule(const std::vector<at::Tensor>& ts,
-                        int64_t chunk_elems, const at::Device& dev) {
+Schedule build_schedule_uncached(const std::vector<at::Tensor>& ts,
+                                 int64_t chunk_elems, const at::Device& dev) {
   std::vector<int> bt;
   std::vector<int64_t> bo;
   for (size_t l = 0; l < ts.size(); ++l) {
@@ -125,6 +127,29 @@ Schedule build_schedule(const std::vector<at::Tensor>& ts,
   s.bo = at::from_blob(bo.data(), {(int64_t)bo.size()},
                        at::TensorOptions().dtype(at::kLong))
              .to(dev, /*non_blocking=*/false);
+  return s;
+}
+
+// The tensor sets of the fused calls are stable across steps (the same
+// parameters every iteration): cache the device-side schedules keyed by
+// the size list + chunking so the per-step host->device copies go away.
+Schedule build_schedule(const std::vector<at::Tensor>& ts,
+                        int64_t chunk_elems, const at::Device& dev) {
+  static std::unordered_map<std::string, Schedule> cache;
+  std::string key;
+  key.reserve(ts.size() * 9 + 16);
+  key += std::to_string(chunk_elems);
+  key += '/';
+  key += std::to_string(dev.index());
+  for (auto& t : ts) {
+    key += ':';
+    key += std::to_string(t.numel());
+  }
+  auto it = cache.find(key);
+  if (it != cache.end()) return it->second;
+  if (cache.size() > 256) cache.clear();  // bound memory; rebuilt on demand
+  auto s = build_schedule_uncached(ts, chunk_elems, dev);
+  cache.emplace(std::move(key), s);
   return s;
 }
 
