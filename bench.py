@@ -27,7 +27,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--batch-size", type=int, default=1024,
+    p.add_argument("--batch-size", type=int, default=2048,
                    help="per-GPU batch size")
     p.add_argument("--arch", default="resnet18")
     p.add_argument("--ts", action="store_true",
