@@ -48,6 +48,11 @@ def main():
     from bdbnn_amd.parallel import init_distributed
 
     rank, local_rank, world_size = init_distributed()
+    if world_size > 1:
+        # per-rank MIOpen find db: concurrent ranks otherwise serialize on
+        # the shared user-db file lock during warmup
+        os.environ.setdefault("MIOPEN_USER_DB_PATH",
+                              f"/tmp/miopen-rank{local_rank}")
     use_cuda = torch.cuda.is_available()
     device = torch.device(f"cuda:{local_rank}") if use_cuda else torch.device("cpu")
     if use_cuda:

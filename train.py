@@ -209,6 +209,9 @@ def _spawned(local_rank, world_size, argv):
 def _run(args):
     rank, local_rank, world_size = init_distributed(
         backend=None if args.dist_backend == "nccl" else args.dist_backend)
+    if world_size > 1:
+        os.environ.setdefault("MIOPEN_USER_DB_PATH",
+                              f"/tmp/miopen-rank{local_rank}")
     setup_logging(args, rank)
     log = logging.getLogger("bdbnn")
     log.info(f"bdbnn_amd {bdbnn_amd.__version__} rank {rank}/{world_size}")
