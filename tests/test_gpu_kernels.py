@@ -515,3 +515,47 @@ def test_fused_cross_entropy(dtype):
     assert torch.allclose(s.grad.float(), s2.grad,
                           atol=1e-5 if dtype == torch.float32 else 1e-3,
                           rtol=1e-2)
+
+
+def test_fused_bn_act_train_bf16():
+    from bdbnn_amd.ops.bn_act import fused_bn_act
+    from bdbnn_amd.ops.activations import ChannelPReLU
+    torch.manual_seed(21)
+    C = 64
+    bn = torch.nn.BatchNorm2d(C).cuda()
+    bn2 = torch.nn.BatchNorm2d(C).cuda()
+    bn2.load_state_dict(bn.state_dict())
+    act = ChannelPReLU(C).cuda()
+    x = _cl(torch.randn(8, C, 14, 14, device="cuda",
+                        dtype=torch.bfloat16) * 2).requires_grad_(True)
+    skip = _cl(torch.randn_like(x.detach())).requires_grad_(True)
+    out = fused_bn_act(x, bn, act, skip=skip)
+    x2 = x.detach().float().clone().requires_grad_(True)
+    s2 = skip.detach().float().clone().requires_grad_(True)
+    z = bn2(x2) + s2
+    ref = torch.nn.functional.prelu(z, act.weight.detach())
+    assert torch.allclose(out.float(), _cl(ref), atol=5e-2, rtol=1e-2)
+    g = torch.randn_like(out)
+    out.backward(g)
+    ref.backward(_cl(g.float()))
+    assert torch.allclose(x.grad.float(), x2.grad, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(skip.grad.float(), s2.grad, atol=5e-2, rtol=5e-2)
+
+
+def test_model_eval_gpu_matches_cpu():
+    from bdbnn_amd.models import imagenet as im
+    torch.manual_seed(22)
+    m_cpu = im.resnet18(False)
+    # populate running stats with a couple of train steps
+    for _ in range(2):
+        m_cpu(torch.randn(4, 3, 64, 64))
+    m_gpu = im.resnet18(False)
+    m_gpu.load_state_dict(m_cpu.state_dict())
+    m_gpu = m_gpu.cuda().to(memory_format=torch.channels_last).eval()
+    m_cpu.eval()
+    x = torch.randn(2, 3, 64, 64)
+    with torch.no_grad():
+        out_g = m_gpu(_cl(x.cuda()))
+        out_c = m_cpu(x)
+    assert torch.allclose(out_g.cpu(), out_c, atol=2e-2, rtol=1e-2), \
+        (out_g.cpu() - out_c).abs().max().item()
