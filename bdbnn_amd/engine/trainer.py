@@ -302,9 +302,40 @@ class Trainer:
                 self.writer.add_scalar("Val Acc5", top5.avg, epoch)
         return top1.avg
 
+    def _emergency_save(self, epoch):
+        """Crash/preemption checkpoint (SIGTERM/KeyboardInterrupt): the
+        reference loses the epoch on any failure (SURVEY.md section 5.3);
+        pair with --auto-resume for unattended recovery."""
+        if self.rank == 0:
+            save_state(self.model, self.optimizer, epoch, self.args.arch,
+                       self.best_acc1, False, self.args.log_path)
+            log.info(f"emergency checkpoint saved at epoch {epoch}")
+
     def fit(self, train_loader, val_loader):
         a = self.args
+        import signal
+
+        def _sigterm(_sig, _frm):
+            raise KeyboardInterrupt
+
+        try:
+            prev_handler = signal.signal(signal.SIGTERM, _sigterm)
+        except ValueError:  # not the main thread
+            prev_handler = None
+        try:
+            return self._fit(train_loader, val_loader)
+        except KeyboardInterrupt:
+            self._emergency_save(self._current_epoch)
+            raise
+        finally:
+            if prev_handler is not None:
+                signal.signal(signal.SIGTERM, prev_handler)
+
+    def _fit(self, train_loader, val_loader):
+        a = self.args
+        self._current_epoch = self.start_epoch
         for epoch in range(self.start_epoch, a.epochs):
+            self._current_epoch = epoch
             self.train_epoch(train_loader, epoch)
             acc1 = self.validate(val_loader, epoch)
             self.scheduler.step()

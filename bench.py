@@ -36,6 +36,8 @@ def parse_args():
                    help="binary inference path: packed weights + hipGraph "
                         "(BASELINE config 5, throughput-only)")
     p.add_argument("--no-kurt", action="store_true")
+    p.add_argument("--diffkurt", action="store_true",
+                   help="per-layer kurtosis target lists (BASELINE config 4)")
     p.add_argument("--image", type=int, default=224)
     return p.parse_args()
 
@@ -73,7 +75,7 @@ def main():
         w_kurtosis_target = 1.8
         w_lambda_kurtosis = 1.0
         kurtosis_mode = "avg"
-        diffkurt = False
+        diffkurt = args.diffkurt
         kurtepoch = 0
         react = False
         alpha = 0.9

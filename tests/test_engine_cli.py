@@ -119,3 +119,14 @@ def test_evaluate_mode():
          "--epochs", "1"], steps=2)
     acc = trainer.validate(loader)
     assert 0.0 <= acc <= 100.0
+
+
+def test_emergency_save_writes_checkpoint(tmp_path):
+    args = _args(["--dataset", "cifar10", "-a", "resnet20", "-b", "4",
+                  "--epochs", "1"])
+    args.log_path = str(tmp_path)
+    model = build_model(args, "resnet20", True)
+    trainer = Trainer(model, args, device=torch.device("cpu"))
+    trainer._emergency_save(epoch=2)
+    import os
+    assert os.path.exists(os.path.join(str(tmp_path), "checkpoint.pth.tar"))
