@@ -30,10 +30,10 @@
 // per block (no integer div/mod in the staging loop).
 #include "common.h"
 
-#define TILE_M 256
+#define TILE_M 128
 #define TILE_K 64
 #define CHUNK 8
-#define A_PER_THREAD 8  // CHUNK*TILE_M / 256
+#define A_PER_THREAD 4  // CHUNK*TILE_M / 256
 #define W_PER_THREAD 2  // CHUNK*TILE_K / 256
 #define MAX_WORDS 160   // KH*KW*CW <= 9*16 (C<=512, 3x3); guarded on host
 
@@ -108,10 +108,10 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
     }
   __syncthreads();
 
-  // thread's register tile: rows r0..r0+7, channels kq..kq+7
-  const int r0 = (tid / 8) * 8;
-  const int kq = (tid % 8) * 8;
-  int acc[8][8] = {};
+  // thread's register tile: rows r0..r0+7, channels kq..kq+3
+  const int r0 = (tid / 16) * 8;
+  const int kq = (tid % 16) * 4;
+  int acc[8][4] = {};
 
   // staging coordinates (fixed per thread)
   int a_c[A_PER_THREAD], a_r[A_PER_THREAD];
@@ -174,16 +174,15 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
     if (more) STAGE_LOAD((ch + 1) * CHUNK);          // issue early
 #pragma unroll
     for (int c = 0; c < CHUNK; ++c) {
-      uint32_t a8[8], b8[8];
+      uint32_t a8[8], b4[4];
       *(uint4*)&a8[0] = *(const uint4*)&a_lds[buf][c][r0];
       *(uint4*)&a8[4] = *(const uint4*)&a_lds[buf][c][r0 + 4];
-      *(uint4*)&b8[0] = *(const uint4*)&w_lds[buf][c][kq];
-      *(uint4*)&b8[4] = *(const uint4*)&w_lds[buf][c][kq + 4];
+      *(uint4*)b4 = *(const uint4*)&w_lds[buf][c][kq];
 #pragma unroll
       for (int i = 0; i < 8; ++i)
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          acc[i][j] += __popc(a8[i] ^ b8[j]);
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] += __popc(a8[i] ^ b4[j]);
     }
     if (more) STAGE_WRITE(buf ^ 1);                  // write late
     __syncthreads();
@@ -192,61 +191,60 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
 #undef STAGE_WRITE
 
   // ---- epilogue: scale, pad-correction, store ----
-  float al[8];
+  float al[4];
 #pragma unroll
-  for (int j = 0; j < 8; ++j) {
+  for (int j = 0; j < 4; ++j) {
     int kg = k0_blk + kq + j;
     al[j] = (kg < p.K) ? alpha[kg] : 0.f;
   }
-  float st1[STATS ? 8 : 1] = {}, st2[STATS ? 8 : 1] = {};
+  float st1[STATS ? 4 : 1] = {}, st2[STATS ? 4 : 1] = {};
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
     int64_t sp = m0 + r0 + i;
     if (sp >= M) continue;
     unsigned inv = row_inv[r0 + i];
-    float corr[8] = {};
+    float corr[4] = {0.f, 0.f, 0.f, 0.f};
     if (inv) {
       for (int t = 0; t < p.T; ++t)
         if ((inv >> t) & 1) {
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
+          for (int j = 0; j < 4; ++j) {
             int kg = k0_blk + kq + j;
             if (kg < p.K) corr[j] += stab[(int64_t)kg * p.T + t];
           }
         }
     }
     if constexpr (sizeof(TO) == 2) {
-      uint16_t vals[8];
+      uint16_t vals[4];
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
+      for (int j = 0; j < 4; ++j)
         vals[j] = f32_to_bf16(al[j] * (2.f * acc[i][j] + p.base - corr[j]));
       int kg = k0_blk + kq;
-      if (kg + 7 < p.K)
-        *(uint4*)&out[sp * p.K + kg] = *(uint4*)vals;
+      if (kg + 3 < p.K)
+        *(uint2*)&out[sp * p.K + kg] = *(uint2*)vals;
       else
-        for (int j = 0; j < 8 && kg + j < p.K; ++j)
+        for (int j = 0; j < 4 && kg + j < p.K; ++j)
           out[sp * p.K + kg + j] = vals[j];
       if constexpr (STATS)
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
+        for (int j = 0; j < 4; ++j) {
           float v = bf16_to_f32(vals[j]);   // stats of the ROUNDED value
           st1[j] += v; st2[j] += v * v;
         }
     } else {
-      float vals[8];
+      float vals[4];
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
+      for (int j = 0; j < 4; ++j)
         vals[j] = al[j] * (2.f * acc[i][j] + p.base - corr[j]);
       int kg = k0_blk + kq;
-      if (kg + 7 < p.K) {
-        *(float4*)&out[sp * p.K + kg] = *(float4*)&vals[0];
-        *(float4*)&out[sp * p.K + kg + 4] = *(float4*)&vals[4];
-      } else
-        for (int j = 0; j < 8 && kg + j < p.K; ++j)
+      if (kg + 3 < p.K)
+        *(float4*)&out[sp * p.K + kg] = *(float4*)vals;
+      else
+        for (int j = 0; j < 4 && kg + j < p.K; ++j)
           ((float*)out)[sp * p.K + kg + j] = vals[j];
       if constexpr (STATS)
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
+        for (int j = 0; j < 4; ++j) {
           st1[j] += vals[j]; st2[j] += vals[j] * vals[j];
         }
     }
@@ -254,7 +252,7 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
   if constexpr (STATS) {
     __syncthreads();  // csum init visible; stores above done per thread
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
+    for (int j = 0; j < 4; ++j) {
       atomicAdd(&csum[0][kq + j], st1[j]);
       atomicAdd(&csum[1][kq + j], st2[j]);
     }
