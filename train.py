@@ -191,8 +191,8 @@ def main(argv=None):
             os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
             os.environ.setdefault("MASTER_PORT", "29513")
             mp.spawn(_spawned, nprocs=ngpus, args=(ngpus, argv))
-            return
-    _run(args)
+            return None
+    return _run(args)
 
 
 def _spawned(local_rank, world_size, argv):
