@@ -56,3 +56,11 @@ def test_cli_resume_round_trip(tiny_synthetic, tmp_path):
     best = _run_cli(tmp_path / "b", ["--resume", ckpts[0],
                                      "--reset_resume"])
     assert best is not None
+
+
+def test_dist_url_sets_rendezvous_env(tiny_synthetic, tmp_path, monkeypatch):
+    monkeypatch.delenv("MASTER_ADDR", raising=False)
+    monkeypatch.delenv("MASTER_PORT", raising=False)
+    _run_cli(tmp_path, ["--dist-url", "tcp://127.0.0.1:23456", "-e"])
+    assert os.environ["MASTER_ADDR"] == "127.0.0.1"
+    assert os.environ["MASTER_PORT"] == "23456"

@@ -183,20 +183,35 @@ def main(argv=None):
     args.log_path = os.path.join(args.log_path,
                                  str(args.w_kurtosis_target), stamp)
 
+    # honor the reference's rendezvous flags when torchrun env is absent
+    if "MASTER_ADDR" not in os.environ:
+        if args.dist_url.startswith("tcp://"):
+            host, _, port = args.dist_url[len("tcp://"):].partition(":")
+            os.environ["MASTER_ADDR"] = host or args.master_addr
+            if port:
+                os.environ.setdefault("MASTER_PORT", port)
+        elif args.master_addr:
+            os.environ["MASTER_ADDR"] = args.master_addr
+
     if args.multiprocessing_distributed and "RANK" not in os.environ:
-        # spawn one process per GPU locally (torchrun does this when used)
+        # spawn one process per GPU; multi-node: --world-size nodes with
+        # --rank node_rank each (the reference's rank math is broken for
+        # this case, SURVEY.md 2.1 — here rank = node_rank*ngpus + gpu)
         ngpus = max(torch.cuda.device_count(), 1)
-        if ngpus > 1:
+        if ngpus > 1 or args.world_size > 1:
             import torch.multiprocessing as mp
             os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
             os.environ.setdefault("MASTER_PORT", "29513")
-            mp.spawn(_spawned, nprocs=ngpus, args=(ngpus, argv))
+            mp.spawn(_spawned,
+                     nprocs=ngpus,
+                     args=(ngpus, args.world_size * ngpus,
+                           args.rank * ngpus, argv))
             return None
     return _run(args)
 
 
-def _spawned(local_rank, world_size, argv):
-    os.environ["RANK"] = str(local_rank)
+def _spawned(local_rank, ngpus, world_size, rank_base, argv):
+    os.environ["RANK"] = str(rank_base + local_rank)
     os.environ["LOCAL_RANK"] = str(local_rank)
     os.environ["WORLD_SIZE"] = str(world_size)
     args = build_parser().parse_args(argv)
