@@ -52,6 +52,8 @@ void bdbnn_ce_fwd(const void*, const int64_t*, float*, float*, int, int,
                   bool, hipStream_t);
 void bdbnn_ce_bwd(const void*, const int64_t*, const float*, void*, float,
                   int, int, bool, hipStream_t);
+void bdbnn_conv_dgrad(const void*, const uint32_t*, const float*, void*,
+                      int, int, int, int, int, int, hipStream_t);
 void bdbnn_prelu_fwd(const void*, const float*, void*, int64_t, int, bool,
                      hipStream_t);
 void bdbnn_prelu_bwd(const void*, const void*, const float*, void*, float*,
@@ -602,6 +604,29 @@ at::Tensor bn_act_eval(const at::Tensor& x,
   return out;
 }
 
+// ---------------- experimental MFMA dgrad ----------------
+
+at::Tensor conv_dgrad(const at::Tensor& g, const at::Tensor& wp,
+                      const at::Tensor& alpha, int64_t C) {
+  TORCH_CHECK(g.is_cuda() && g.dim() == 4 &&
+                  g.scalar_type() == at::kBFloat16 &&
+                  g.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "conv_dgrad: bf16 channels_last grad");
+  int N = (int)g.size(0), K = (int)g.size(1);
+  int H = (int)g.size(2), W = (int)g.size(3);
+  TORCH_CHECK(wp.size(0) == K && wp.size(1) == 3 && wp.size(2) == 3,
+              "conv_dgrad: packed 3x3 weights");
+  TORCH_CHECK(((int64_t)N * H * W) % 128 == 0 && C % 64 == 0 && K % 16 == 0,
+              "conv_dgrad: shape constraints");
+  int CW = (int)wp.size(3);
+  auto dx = at::empty({N, C, H, W}, g.options(),
+                      at::MemoryFormat::ChannelsLast);
+  bdbnn_conv_dgrad(g.data_ptr(), (const uint32_t*)wp.data_ptr<int>(),
+                   alpha.data_ptr<float>(), dx.data_ptr(), N, H, W, (int)C,
+                   K, CW, cur_stream());
+  return dx;
+}
+
 // ---------------- kurtosis ----------------
 
 std::vector<at::Tensor> kurtosis_fwd(const std::vector<at::Tensor>& ws,
@@ -735,6 +760,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused BN(+add)(+act) training forward");
   m.def("bn_act_bwd", &bn_act_bwd, "fused BN(+add)(+act) backward");
   m.def("bn_act_eval", &bn_act_eval, "fused BN(+add)(+act) eval forward");
+  m.def("conv_dgrad", &conv_dgrad,
+        "EXPERIMENTAL MFMA bf16 dgrad (3x3/s1/p1, packed weights)");
   m.def("kurtosis_fwd", &kurtosis_fwd, "fused multi-tensor kurtosis fwd");
   m.def("kurtosis_bwd", &kurtosis_bwd, "fused multi-tensor kurtosis bwd");
   m.def("weight_kd_fwd", &weight_kd_fwd, "fused weight-space KD fwd");

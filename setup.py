@@ -25,6 +25,7 @@ sources = [os.path.join(CSRC, f) for f in (
     "bn_act.hip",
     "pool.hip",
     "loss.hip",
+    "conv_dgrad.hip",
     "xnor_conv.hip",
     "kurtosis.hip",
     "kd.hip",
@@ -39,7 +40,11 @@ setup(
             sources=sources,
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
-                "nvcc": ["-O3", "-std=c++17"],
+                # rocWMMA (conv_dgrad.hip) needs HIP half conversions,
+                # which torch's extension defaults disable
+                "nvcc": ["-O3", "-std=c++17",
+                         "-U__HIP_NO_HALF_CONVERSIONS__",
+                         "-U__HIP_NO_HALF_OPERATORS__"],
             },
         )
     ],

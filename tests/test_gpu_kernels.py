@@ -559,3 +559,29 @@ def test_model_eval_gpu_matches_cpu():
         out_c = m_cpu(x)
     assert torch.allclose(out_g.cpu(), out_c, atol=2e-2, rtol=1e-2), \
         (out_g.cpu() - out_c).abs().max().item()
+
+
+# ---------------- experimental MFMA dgrad (round-2 seed) ----------------
+
+import os as _os
+
+
+@pytest.mark.skipif(_os.environ.get("BDBNN_EXPERIMENTAL") != "1",
+                    reason="experimental kernel; enable with BDBNN_EXPERIMENTAL=1")
+def test_experimental_conv_dgrad_matches_reference():
+    torch.manual_seed(23)
+    nat = _nat()
+    for (N, C, H, K) in [(2, 64, 8, 64), (1, 128, 16, 32)]:
+        g = torch.randn(N, K, H, H, device="cuda", dtype=torch.bfloat16)
+        g = _cl(g)
+        w = torch.randn(K, C, 3, 3, device="cuda")
+        wp, alpha, stab = nat.weight_pack(w)
+        if (N * H * H) % 128 or C % 64 or K % 16:
+            continue
+        dx = nat.conv_dgrad(g, wp, alpha, C)
+        wb = (weight_scale(w) * binsign(w)).to(torch.bfloat16)
+        ref = torch.nn.functional.conv_transpose2d(
+            g.float(), wb.float(), None, stride=1, padding=1)
+        assert dx.shape == ref.shape
+        assert torch.allclose(dx.float(), _cl(ref), atol=0.5, rtol=2e-2), \
+            (dx.float() - _cl(ref)).abs().max().item()
