@@ -54,6 +54,8 @@ void bdbnn_ce_bwd(const void*, const int64_t*, const float*, void*, float,
                   int, int, bool, hipStream_t);
 void bdbnn_conv_dgrad(const void*, const uint32_t*, const float*, void*,
                       int, int, int, int, int, int, hipStream_t);
+void bdbnn_conv_wgrad(const void*, const uint32_t*, float*, int, int, int,
+                      int, int, int, hipStream_t);
 void bdbnn_prelu_fwd(const void*, const float*, void*, int64_t, int, bool,
                      hipStream_t);
 void bdbnn_prelu_bwd(const void*, const void*, const float*, void*, float*,
@@ -627,6 +629,26 @@ at::Tensor conv_dgrad(const at::Tensor& g, const at::Tensor& wp,
   return dx;
 }
 
+at::Tensor conv_wgrad(const at::Tensor& g, const at::Tensor& xp,
+                      int64_t C) {
+  TORCH_CHECK(g.is_cuda() && g.dim() == 4 &&
+                  g.scalar_type() == at::kBFloat16 &&
+                  g.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "conv_wgrad: bf16 channels_last grad");
+  int N = (int)g.size(0), K = (int)g.size(1);
+  int H = (int)g.size(2), W = (int)g.size(3);
+  TORCH_CHECK(xp.size(0) == N && xp.size(1) == H && xp.size(2) == W,
+              "conv_wgrad: packed activations shape");
+  TORCH_CHECK(((int64_t)N * H * W) % 16 == 0 && K % 32 == 0 && C % 32 == 0,
+              "conv_wgrad: shape constraints");
+  int CW = (int)xp.size(3);
+  auto dw = at::zeros({K, C, 3, 3}, g.options().dtype(at::kFloat));
+  bdbnn_conv_wgrad(g.data_ptr(), (const uint32_t*)xp.data_ptr<int>(),
+                   dw.data_ptr<float>(), N, H, W, (int)C, K, CW,
+                   cur_stream());
+  return dw;
+}
+
 // ---------------- kurtosis ----------------
 
 std::vector<at::Tensor> kurtosis_fwd(const std::vector<at::Tensor>& ws,
@@ -762,6 +784,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_act_eval", &bn_act_eval, "fused BN(+add)(+act) eval forward");
   m.def("conv_dgrad", &conv_dgrad,
         "EXPERIMENTAL MFMA bf16 dgrad (3x3/s1/p1, packed weights)");
+  m.def("conv_wgrad", &conv_wgrad,
+        "EXPERIMENTAL MFMA bf16 wgrad (3x3/s1/p1, packed activations)");
   m.def("kurtosis_fwd", &kurtosis_fwd, "fused multi-tensor kurtosis fwd");
   m.def("kurtosis_bwd", &kurtosis_bwd, "fused multi-tensor kurtosis bwd");
   m.def("weight_kd_fwd", &weight_kd_fwd, "fused weight-space KD fwd");
