@@ -587,9 +587,9 @@ def test_experimental_conv_dgrad_matches_reference():
             (dx.float() - _cl(ref)).abs().max().item()
 
 
-@pytest.mark.skipif(_os.environ.get("BDBNN_EXPERIMENTAL") != "1",
-                    reason="experimental kernel; enable with BDBNN_EXPERIMENTAL=1")
 def test_experimental_conv_wgrad_matches_reference():
+    # validated on MI355X (first-shot numerics pass); default-off in the
+    # training path until round-2 perf tuning
     torch.manual_seed(24)
     nat = _nat()
     for (N, C, H, K) in [(2, 32, 8, 32), (1, 64, 16, 64)]:
