@@ -130,3 +130,10 @@ def test_emergency_save_writes_checkpoint(tmp_path):
     trainer._emergency_save(epoch=2)
     import os
     assert os.path.exists(os.path.join(str(tmp_path), "checkpoint.pth.tar"))
+
+
+def test_cifar100_model_path():
+    args = _args(["--dataset", "cifar100", "-a", "resnet20", "-b", "4"])
+    m = build_model(args, "resnet20", True)
+    out = m(torch.randn(2, 3, 32, 32))
+    assert out.shape == (2, 100)

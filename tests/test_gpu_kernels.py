@@ -607,3 +607,18 @@ def test_experimental_conv_wgrad_matches_reference():
         assert dw.shape == ref.shape
         assert torch.allclose(dw, ref, atol=0.5, rtol=2e-2), \
             (dw - ref).abs().max().item()
+
+
+def test_vgg_small_gpu_step():
+    from bdbnn_amd.models.cifar10 import vgg_small
+    from bdbnn_amd.ops.optim import FusedAdam
+    torch.manual_seed(25)
+    m = vgg_small().cuda().to(memory_format=torch.channels_last)
+    opt = FusedAdam(m.parameters(), lr=1e-3)
+    x = _cl(torch.randn(8, 3, 32, 32, device="cuda"))
+    y = torch.randint(0, 10, (8,), device="cuda")
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        loss = torch.nn.functional.cross_entropy(m(x), y)
+    loss.backward()
+    opt.step()
+    assert torch.isfinite(loss)
