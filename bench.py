@@ -62,6 +62,11 @@ def main():
 
     torch.manual_seed(1234 + rank)
 
+    if not use_cuda and args.batch_size >= 512:
+        # CPU-only machine (no-flag contract: must finish in minutes)
+        args.batch_size = 16
+        args.image = min(args.image, 64)
+
     class TArgs:  # minimal trainer config (mirrors train.py flags)
         arch = args.arch
         dataset = "imagenet"
