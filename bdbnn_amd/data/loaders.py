@@ -16,6 +16,7 @@ Augmentations (random crop + horizontal flip + normalize) are
 implemented as tensor ops inside the datasets.
 """
 
+import logging
 import os
 import pickle
 
@@ -23,6 +24,8 @@ import numpy as np
 import torch
 from torch.utils.data import DataLoader, Dataset
 from torch.utils.data.distributed import DistributedSampler
+
+_LOG = logging.getLogger("bdbnn")
 
 _CIFAR_MEAN = (0.4914, 0.4822, 0.4465)
 _CIFAR_STD = (0.2023, 0.1994, 0.2010)
@@ -113,6 +116,39 @@ class CIFAR100Dataset(CIFARBase):
         return ["train"] if train else ["test"]
 
 
+def _random_resized_crop_box(width, height, scale=(0.08, 1.0),
+                             ratio=(3.0 / 4.0, 4.0 / 3.0)):
+    """Sample a crop box with torchvision RandomResizedCrop semantics
+    (ref:loader.py:59-64 uses transforms.RandomResizedCrop(224)): area
+    fraction ~ U(scale), log-aspect ~ U(log(ratio)); 10 attempts, then a
+    center fallback clamped to the valid ratio range.
+
+    Returns (left, top, w, h).
+    """
+    import math
+    area = width * height
+    log_ratio = (math.log(ratio[0]), math.log(ratio[1]))
+    for _ in range(10):
+        target_area = area * float(
+            torch.empty(1).uniform_(scale[0], scale[1]))
+        aspect = math.exp(float(torch.empty(1).uniform_(*log_ratio)))
+        w = int(round((target_area * aspect) ** 0.5))
+        h = int(round((target_area / aspect) ** 0.5))
+        if 0 < w <= width and 0 < h <= height:
+            left = int(torch.randint(0, width - w + 1, (1,)))
+            top = int(torch.randint(0, height - h + 1, (1,)))
+            return left, top, w, h
+    # fallback: largest center crop within the ratio range
+    in_ratio = width / height
+    if in_ratio < ratio[0]:
+        w, h = width, int(round(width / ratio[0]))
+    elif in_ratio > ratio[1]:
+        w, h = int(round(height * ratio[1])), height
+    else:
+        w, h = width, height
+    return (width - w) // 2, (height - h) // 2, w, h
+
+
 class ImageFolderDataset(Dataset):
     """Minimal ImageNet-style folder reader (requires PIL)."""
 
@@ -141,14 +177,25 @@ class ImageFolderDataset(Dataset):
         path, label = self.samples[idx]
         img = Image.open(path).convert("RGB")
         if self.train:
-            img = img.resize((224, 224))
+            # RandomResizedCrop(224) + HFlip (ref:loader.py:59-64)
+            left, top, w, h = _random_resized_crop_box(*img.size)
+            img = img.resize((224, 224), box=(left, top, left + w, top + h))
             t = torch.from_numpy(np.asarray(img, np.float32) / 255.0).permute(2, 0, 1)
             if torch.rand(()) < 0.5:
                 t = t.flip(-1)
         else:
-            img = img.resize((256, 256))
+            # aspect-preserving Resize(256) + CenterCrop(224)
+            # (ref:loader.py:75-80)
+            wd, ht = img.size
+            if wd <= ht:
+                nw, nh = 256, max(1, int(round(256 * ht / wd)))
+            else:
+                nw, nh = max(1, int(round(256 * wd / ht))), 256
+            img = img.resize((nw, nh))
             t = torch.from_numpy(np.asarray(img, np.float32) / 255.0).permute(2, 0, 1)
-            t = t[:, 16:240, 16:240]
+            i0 = (nh - 224) // 2
+            j0 = (nw - 224) // 2
+            t = t[:, i0:i0 + 224, j0:j0 + 224]
         return _normalize(t, _IMAGENET_MEAN, _IMAGENET_STD), label
 
 
@@ -164,11 +211,34 @@ def _make_loader(dataset, batch_size, shuffle, workers, distributed,
                       persistent_workers=workers > 0)
 
 
+def _resolve_dataset_dir(data_path, subdir, synthetic, what):
+    """Fallback policy: synthetic data is used only when asked for
+    (``synthetic=True``) or when no path was given at all (then with a
+    loud warning).  A missing EXPLICIT path raises — a run pointed at a
+    wrong directory must not silently 'train' on noise.
+    """
+    if synthetic:
+        return None
+    if data_path is None:
+        _LOG.warning(
+            "no data path given for %s — falling back to SYNTHETIC random "
+            "data (pass --synthetic-data to silence, or a dataset root)",
+            what)
+        return None
+    full = os.path.join(data_path, subdir) if subdir else data_path
+    if not os.path.isdir(full):
+        raise FileNotFoundError(
+            f"{what}: dataset directory {full!r} does not exist "
+            "(pass --synthetic-data for synthetic random data)")
+    return full
+
+
 def dataloader_cifar10(split="train", batch_size=128, data_path=None,
                        distributed=False, workers=4, synthetic=False):
     train = split == "train"
-    if synthetic or data_path is None or not os.path.isdir(
-            os.path.join(data_path or "", "cifar-10-batches-py")):
+    found = _resolve_dataset_dir(data_path, "cifar-10-batches-py",
+                                 synthetic, "cifar10")
+    if found is None:
         ds = SyntheticImageDataset(50000 if train else 10000, (3, 32, 32), 10)
     else:
         ds = CIFAR10Dataset(data_path, train=train)
@@ -178,20 +248,22 @@ def dataloader_cifar10(split="train", batch_size=128, data_path=None,
 def dataloader_cifar100(split="train", batch_size=128, data_path=None,
                         distributed=False, workers=4, synthetic=False):
     train = split == "train"
-    if synthetic or data_path is None or not os.path.isdir(
-            os.path.join(data_path or "", "cifar-100-python")):
+    found = _resolve_dataset_dir(data_path, "cifar-100-python",
+                                 synthetic, "cifar100")
+    if found is None:
         ds = SyntheticImageDataset(50000 if train else 10000, (3, 32, 32), 100)
     else:
         ds = CIFAR100Dataset(data_path, train=train)
     return _make_loader(ds, batch_size, train, workers, distributed)
 
 
-def dataloader_imagenet(split="train", batch_size=128, data_path="./",
+def dataloader_imagenet(split="train", batch_size=128, data_path=None,
                         distributed=False, workers=8, synthetic=False,
                         synthetic_len=100000):
     train = split == "train"
-    sub = os.path.join(data_path, "train" if train else "val")
-    if synthetic or not os.path.isdir(sub):
+    sub = _resolve_dataset_dir(data_path, "train" if train else "val",
+                               synthetic, "imagenet")
+    if sub is None:
         ds = SyntheticImageDataset(synthetic_len if train else 10000,
                                    (3, 224, 224), 1000)
     else:

@@ -56,12 +56,45 @@ def test_dataloader_cifar10_real_files(tmp_path):
     assert x.shape == (8, 3, 32, 32)
 
 
-def test_dataloader_falls_back_to_synthetic(tmp_path):
+def test_dataloader_missing_explicit_path_raises(tmp_path):
+    # a wrong --data path must not silently 'train' on noise
+    import pytest
+    with pytest.raises(FileNotFoundError):
+        dataloader_imagenet("train", batch_size=4,
+                            data_path=str(tmp_path / "nope"), workers=0)
+
+
+def test_dataloader_synthetic_when_requested(tmp_path):
     loader = dataloader_imagenet("train", batch_size=4,
                                  data_path=str(tmp_path), workers=0,
-                                 synthetic_len=16)
+                                 synthetic=True, synthetic_len=16)
     x, y = next(iter(loader))
     assert x.shape == (4, 3, 224, 224)
+
+
+def test_dataloader_no_path_falls_back_with_warning(caplog):
+    import logging
+    with caplog.at_level(logging.WARNING, logger="bdbnn"):
+        loader = dataloader_imagenet("train", batch_size=4, data_path=None,
+                                     workers=0, synthetic_len=16)
+    assert any("SYNTHETIC" in r.message for r in caplog.records)
+    x, y = next(iter(loader))
+    assert x.shape == (4, 3, 224, 224)
+
+
+def test_random_resized_crop_box_semantics():
+    # scale in [0.08, 1.0] x area, aspect in [3/4, 4/3] (ref:loader.py:59-64)
+    from bdbnn_amd.data.loaders import _random_resized_crop_box
+    torch.manual_seed(0)
+    areas, aspects = [], []
+    for _ in range(300):
+        left, top, w, h = _random_resized_crop_box(500, 400)
+        assert 0 <= left <= 500 - w and 0 <= top <= 400 - h
+        areas.append(w * h / (500 * 400))
+        aspects.append(w / h)
+    assert min(areas) < 0.25 and max(areas) > 0.5  # spans the scale range
+    assert all(0.70 <= a <= 1.40 for a in aspects)  # rounding slack
+    assert min(aspects) < 0.85 and max(aspects) > 1.15
 
 
 def test_synthetic_loader_drop_last():
