@@ -82,12 +82,13 @@ void bdbnn_kurtosis_fwd(const TensorListArg*, const int*, const int64_t*,
                         hipStream_t);
 void bdbnn_kurtosis_bwd(const TensorListArg*, const PtrList*, const int*,
                         const int64_t*, int, const float*, const float*,
-                        float, hipStream_t);
+                        const float*, hipStream_t);
 void bdbnn_weight_kd_fwd(const TensorListArg*, const PtrList*, const PtrList*,
                          const int*, const int64_t*, int, double*,
                          hipStream_t);
 void bdbnn_weight_kd_bwd(const TensorListArg*, const PtrList*, const PtrList*,
-                         const int*, const int64_t*, int, float, hipStream_t);
+                         const int*, const int64_t*, int, const float*,
+                         hipStream_t);
 void bdbnn_fused_sgd(const TensorListArg*, const PtrList*, const PtrList*,
                      const PtrList*, const int*, const int64_t*, int, float,
                      float, float, hipStream_t);
@@ -674,7 +675,7 @@ std::vector<at::Tensor> kurtosis_fwd(const std::vector<at::Tensor>& ws,
 std::vector<at::Tensor> kurtosis_bwd(const std::vector<at::Tensor>& ws,
                                      const at::Tensor& stats,
                                      const at::Tensor& targets,
-                                     double gscale) {
+                                     const at::Tensor& gscale) {
   auto meta = make_meta(ws);
   auto dev = ws[0].device();
   auto sched = build_schedule(ws, 32 * 1024, dev);
@@ -685,10 +686,12 @@ std::vector<at::Tensor> kurtosis_bwd(const std::vector<at::Tensor>& ws,
   auto gp = make_ptrs(grads);
   auto tgt = targets.to(dev, at::kFloat).contiguous();
   auto st = stats.contiguous();
+  // gscale is a 0-dim DEVICE tensor: no host sync in the backward
+  auto gs = gscale.to(dev, at::kFloat).contiguous();
   bdbnn_kurtosis_bwd(&meta, &gp, sched.bt.data_ptr<int>(),
                      sched.bo.data_ptr<int64_t>(), sched.n_blocks,
                      st.data_ptr<float>(), tgt.data_ptr<float>(),
-                     (float)gscale, cur_stream());
+                     gs.data_ptr<float>(), cur_stream());
   return grads;
 }
 
@@ -710,7 +713,7 @@ at::Tensor weight_kd_fwd(const std::vector<at::Tensor>& ws,
 }
 
 std::vector<at::Tensor> weight_kd_bwd(const std::vector<at::Tensor>& wt,
-                                      double gscale) {
+                                      const at::Tensor& gscale) {
   auto meta = make_meta(wt);
   auto sched = build_schedule(wt, 32 * 1024, wt[0].device());
   auto t_ptr = make_ptrs(wt);
@@ -719,9 +722,10 @@ std::vector<at::Tensor> weight_kd_bwd(const std::vector<at::Tensor>& wt,
     grads.push_back(at::empty_like(t, t.options(),
                                    at::MemoryFormat::Preserve));
   auto gp = make_ptrs(grads);
+  auto gs = gscale.to(wt[0].device(), at::kFloat).contiguous();
   bdbnn_weight_kd_bwd(&meta, &t_ptr, &gp, sched.bt.data_ptr<int>(),
                       sched.bo.data_ptr<int64_t>(), sched.n_blocks,
-                      (float)gscale, cur_stream());
+                      gs.data_ptr<float>(), cur_stream());
   return grads;
 }
 

@@ -41,14 +41,14 @@ __global__ void weight_kd_bwd_kernel(TensorListArg wt_meta, PtrList2 wt,
                                      PtrList2 grads,
                                      const int* __restrict__ block_tensor,
                                      const int64_t* __restrict__ block_off,
-                                     float gscale) {
+                                     const float* __restrict__ gscale) {
   int l = block_tensor[blockIdx.x];
   int64_t n = wt_meta.numel[l];
   int64_t off = block_off[blockIdx.x];
   int64_t end = bd_min(n, off + KD_CHUNK_ELEMS);
   const float* T = wt.ptr[l];
   float* G = grads.ptr[l];
-  float coef = -gscale / (float)n;
+  float coef = -gscale[0] / (float)n;
   for (int64_t i = off + threadIdx.x; i < end; i += blockDim.x)
     G[i] = coef * expf(T[i]);
 }
@@ -66,7 +66,7 @@ extern "C" void bdbnn_weight_kd_fwd(const TensorListArg* meta,
 extern "C" void bdbnn_weight_kd_bwd(const TensorListArg* meta,
                                     const PtrList2* wt, const PtrList2* grads,
                                     const int* bt, const int64_t* bo,
-                                    int n_blocks, float gscale,
+                                    int n_blocks, const float* gscale,
                                     hipStream_t stream) {
   weight_kd_bwd_kernel<<<n_blocks, 256, 0, stream>>>(
       *meta, *wt, *grads, bt, bo, gscale);

@@ -82,14 +82,14 @@ __global__ void kurt_bwd_kernel(TensorListArg lists, GradPtrs gp,
                                 const int64_t* __restrict__ block_off,
                                 const float* __restrict__ stats,
                                 const float* __restrict__ targets,
-                                float gscale) {
+                                const float* __restrict__ gscale) {
   int l = block_tensor[blockIdx.x];
   const float* w = lists.ptr[l];
   float* g = gp.ptr[l];
   int64_t n = lists.numel[l];
   float mu = stats[l * 4 + 0], sigma = stats[l * 4 + 1];
   float kurt = stats[l * 4 + 2], z3m = stats[l * 4 + 3];
-  float coef = gscale * 2.f * (kurt - targets[l]) * 4.f / (float(n) * sigma);
+  float coef = gscale[0] * 2.f * (kurt - targets[l]) * 4.f / (float(n) * sigma);
   float knn = kurt * float(n) / float(n - 1);
   int64_t off = block_off[blockIdx.x];
   int64_t end = bd_min(n, off + KURT_CHUNK_ELEMS);
@@ -120,7 +120,7 @@ extern "C" void bdbnn_kurtosis_bwd(const TensorListArg* lists,
                                    const int* block_tensor_dev,
                                    const int64_t* block_off_dev,
                                    int n_blocks, const float* stats,
-                                   const float* targets, float gscale,
+                                   const float* targets, const float* gscale,
                                    hipStream_t stream) {
   kurt_bwd_kernel<<<n_blocks, 256, 0, stream>>>(
       *lists, *gp, block_tensor_dev, block_off_dev, stats, targets, gscale);

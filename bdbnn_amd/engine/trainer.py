@@ -32,9 +32,13 @@ from .checkpoint import save_state, load_state
 log = logging.getLogger("bdbnn")
 
 # Hardcoded per-layer kurtosis targets (ref:train.py:466-475,585-589).
+# The teacher-student path uses its OWN 19-entry list
+# (ref:train.py:586-589) — different from the plain-imagenet one.
 DIFFKURT_TARGETS = {
     "imagenet": [1.8, 1.4, 1.4, 1.4, 1.4, 1.2, 1.4, 1.2, 1.2, 1.4, 1.4,
                  1.4, 1.2, 1.2, 1.2, 1.2, 1.4, 1, 1],
+    "imagenet_ts": [1.8, 1.8, 1.8, 1.8, 1.8, 1.8, 1.4, 1.8, 1.8, 1.8,
+                    1.4, 1.4, 1.4, 1.4, 1.8, 1.2, 1.4, 1.2, 1.2],
     "cifar": [1.4] * 14 + [1.8] * 4 + [2.2],
 }
 
@@ -81,7 +85,11 @@ def build_kurtosis_table(model, args):
 
 def kurtosis_targets(args, n_layers):
     if getattr(args, "diffkurt", False):
-        key = "imagenet" if args.dataset == "imagenet" else "cifar"
+        ts = getattr(args, "imagenet_setting_step_2_ts", False)
+        if args.dataset == "imagenet":
+            key = "imagenet_ts" if ts else "imagenet"
+        else:
+            key = "cifar"
         tgt = list(DIFFKURT_TARGETS[key])
         if len(tgt) < n_layers:
             tgt = tgt + [tgt[-1]] * (n_layers - len(tgt))
@@ -229,6 +237,26 @@ class Trainer:
             log, prefix=f"Epoch: [{epoch}]")
 
         ts = self.teacher is not None
+
+        # Device-side meter accumulation: per-step stats are summed into
+        # ONE device tensor and read back only at print_freq / epoch end
+        # (the reference's four .item() calls per step are host syncs
+        # that would also serialize backward against the all-reduce at
+        # DP>1; ref:train.py:518-524).
+        # layout: [loss*n, ce*n, kurt*n, correct1, correct5]
+        dev_acc = torch.zeros(5, dtype=torch.float64, device=self.device)
+        seen = 0        # sample count (host-side; batch sizes are known)
+        kurt_on = False
+
+        def _sync_meters(last_step=None):
+            vals = dev_acc.tolist()        # the one host sync
+            losses.set_totals(vals[0], seen)
+            losses_ce.set_totals(vals[1], seen)
+            if kurt_on:
+                losses_kurt.set_totals(vals[2], seen)
+            top1.set_totals(100.0 * vals[3], seen)
+            top5.set_totals(100.0 * vals[4], seen)
+
         end = time.time()
         for i, (images, target) in enumerate(loader):
             data_time.update(time.time() - end)
@@ -251,19 +279,24 @@ class Trainer:
             with trace_range("optimizer"):
                 self.optimizer.step()
 
-            acc1, acc5 = utils.accuracy(output, target, topk=(1, 5))
             n = images.size(0)
-            losses.update(total.item(), n)
-            losses_ce.update(ce.item(), n)
-            if kurt is not None:
-                losses_kurt.update(kurt.item(), n)
-            top1.update(acc1[0].item(), n)
-            top5.update(acc5[0].item(), n)
+            with torch.no_grad():
+                ck = utils.correct_counts(output, target, (1, 5))
+                kz = kurt.detach() if kurt is not None else \
+                    total.new_zeros(())
+                kurt_on = kurt_on or kurt is not None
+                step_vec = torch.cat(
+                    (torch.stack((total.detach(), ce.detach(), kz)) * n,
+                     ck)).double()
+                dev_acc += step_vec
+            seen += n
 
             batch_time.update(time.time() - end)
             end = time.time()
             if i % a.print_freq == 0 and self.rank == 0:
+                _sync_meters()
                 progress.display(i)
+        _sync_meters()
         for m in (losses, top1, top5):
             m.all_reduce(self.device)
         if self.writer is not None:
@@ -278,6 +311,8 @@ class Trainer:
         losses = utils.AverageMeter("Loss", ":.4e")
         top1 = utils.AverageMeter("Acc@1", ":6.2f")
         top5 = utils.AverageMeter("Acc@5", ":6.2f")
+        dev_acc = None
+        seen = 0
         for images, target in loader:
             images = images.to(self.device, non_blocking=True)
             target = target.to(self.device, non_blocking=True)
@@ -286,11 +321,16 @@ class Trainer:
             with torch.autocast("cuda", dtype=torch.bfloat16, enabled=self.amp):
                 output = self.model(images)
                 loss = self.criterion(output, target)
-            acc1, acc5 = utils.accuracy(output, target, topk=(1, 5))
             n = images.size(0)
-            losses.update(loss.item(), n)
-            top1.update(acc1[0].item(), n)
-            top5.update(acc5[0].item(), n)
+            ck = utils.correct_counts(output, target, (1, 5))
+            step_vec = torch.cat((loss.reshape(1) * n, ck)).double()
+            dev_acc = step_vec if dev_acc is None else dev_acc + step_vec
+            seen += n
+        if dev_acc is not None:
+            vals = dev_acc.tolist()      # one host sync per epoch
+            losses.set_totals(vals[0], seen)
+            top1.set_totals(100.0 * vals[1], seen)
+            top5.set_totals(100.0 * vals[2], seen)
         for m in (losses, top1, top5):
             m.all_reduce(self.device)
         if self.rank == 0:

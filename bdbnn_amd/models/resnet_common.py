@@ -180,12 +180,24 @@ def _resnet(layers, binary, conv_cls=None, act="prelu", **kw):
 
 # ---------------- ImageNet constructors ----------------
 
+def _warn_no_pretrained(name, pretrained):
+    if pretrained:
+        import logging
+        logging.getLogger("bdbnn").warning(
+            "%s(pretrained=True): no pretrained weights ship in this "
+            "offline image — the model is RANDOM-INIT; load real weights "
+            "with --resume_teacher <ckpt> (the reference loads a "
+            "pretrained torchvision teacher, ref:train.py:252-257)", name)
+
+
 def resnet18_real(pretrained=False, num_classes=1000, stem="imagenet"):
+    _warn_no_pretrained("resnet18_real", pretrained)
     return _resnet([2, 2, 2, 2], binary=False, num_classes=num_classes,
                    stem=stem)
 
 
 def resnet34_real(pretrained=False, num_classes=1000, stem="imagenet"):
+    _warn_no_pretrained("resnet34_real", pretrained)
     return _resnet([3, 4, 6, 3], binary=False, num_classes=num_classes,
                    stem=stem)
 

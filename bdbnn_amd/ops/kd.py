@@ -139,7 +139,8 @@ class _FusedWeightKD(torch.autograd.Function):
         grads = []
         if ctx.native:
             nat = _C.native_required()
-            grads = list(nat.weight_kd_bwd(list(wt), float(g)))
+            # g stays a device scalar — no host sync mid-backward
+            grads = list(nat.weight_kd_bwd(list(wt), g.detach()))
         else:
             for b in wt:
                 grads.append(-g * torch.exp(b) / b.numel())

@@ -113,7 +113,10 @@ class _FusedKurtosis(torch.autograd.Function):
             targets, stats = ctx.saved_tensors[:2]
             tensors = ctx.saved_tensors[2:]
             nat = _C.native_required()
-            grads = nat.kurtosis_bwd(list(tensors), stats, targets, float(g))
+            # g stays a device scalar — no host sync mid-backward (it
+            # would serialize backward against the DP all-reduce)
+            grads = nat.kurtosis_bwd(list(tensors), stats, targets,
+                                     g.detach())
             return (None, *grads)
         targets = ctx.saved_tensors[0]
         tensors = ctx.saved_tensors[1:]

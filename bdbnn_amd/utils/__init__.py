@@ -5,6 +5,7 @@ from .utils import (
     AverageMeter,
     ProgressMeter,
     accuracy,
+    correct_counts,
 )
 from ..ops.kd import (
     DistributionLoss,

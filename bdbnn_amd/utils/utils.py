@@ -64,6 +64,15 @@ class AverageMeter:
         self.count += n
         self.avg = self.sum / max(self.count, 1)
 
+    def set_totals(self, total, count, last=None):
+        """Load the meter from an externally accumulated (sum, count) —
+        used by the device-side meter accumulation path, which syncs to
+        host only at print_freq/epoch boundaries."""
+        self.sum = float(total)
+        self.count = int(count)
+        self.avg = self.sum / max(self.count, 1)
+        self.val = self.avg if last is None else float(last)
+
     def get_avg(self):
         return self.avg
 
@@ -97,6 +106,19 @@ class ProgressMeter:
         entries = [self.prefix + self.fmt.format(batch)]
         entries += [str(m) for m in self.meters]
         self.logger.info("\t".join(entries))
+
+
+def correct_counts(output, target, topk=(1, 5)):
+    """Raw top-k correct counts as ONE device tensor [len(topk)] — no
+    host sync; the trainer accumulates these device-side and reads them
+    back only at print_freq/epoch end (the reference syncs 4x per step,
+    ref:train.py:518-524)."""
+    with torch.no_grad():
+        maxk = max(topk)
+        _, pred = output.topk(maxk, dim=1, largest=True, sorted=True)
+        correct = pred.eq(target.view(-1, 1))
+        return torch.stack(
+            [correct[:, :k].sum(dtype=torch.float32) for k in topk])
 
 
 def accuracy(output, target, topk=(1,)):

@@ -252,11 +252,21 @@ def _run(args):
         if os.path.isfile(auto_ckpt):
             log.info(f"auto-resume from {auto_ckpt}")
             trainer.resume(auto_ckpt)
-    if args.resume_teacher and teacher is not None and \
-            os.path.isfile(args.resume_teacher):
-        from bdbnn_amd.engine.checkpoint import load_state
-        load_state(args.resume_teacher, trainer.teacher,
-                   map_location=str(device))
+    if teacher is not None:
+        if args.resume_teacher:
+            if not os.path.isfile(args.resume_teacher):
+                raise SystemExit(
+                    f"--resume_teacher {args.resume_teacher!r}: no such "
+                    "file (a TS run distilling from a random-init teacher "
+                    "is almost certainly not what you want)")
+            from bdbnn_amd.engine.checkpoint import load_state
+            load_state(args.resume_teacher, trainer.teacher,
+                       map_location=str(device))
+        else:
+            log.warning(
+                "teacher-student run without --resume_teacher: the "
+                "teacher is RANDOM-INIT (no pretrained weights in this "
+                "offline image) — KD will distill noise")
 
     train_loader, val_loader = make_loaders(args, world_size > 1)
 
