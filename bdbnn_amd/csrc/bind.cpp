@@ -54,6 +54,10 @@ void bdbnn_ce_bwd(const void*, const int64_t*, const float*, void*, float,
                   int, int, bool, hipStream_t);
 void bdbnn_conv_dgrad(const void*, const uint32_t*, const float*, void*,
                       int, int, int, int, int, int, hipStream_t);
+int bdbnn_conv_dgrad2(const void*, const void*, const uint32_t*, void*,
+                      int, int, int, int, int, hipStream_t);
+void bdbnn_dgrad_wdec(const uint32_t*, const float*, void*, int, int,
+                      hipStream_t);
 void bdbnn_conv_wgrad(const void*, const uint32_t*, float*, int, int, int,
                       int, int, int, hipStream_t);
 void bdbnn_prelu_fwd(const void*, const float*, void*, int64_t, int, bool,
@@ -607,7 +611,56 @@ at::Tensor bn_act_eval(const at::Tensor& x,
   return out;
 }
 
-// ---------------- experimental MFMA dgrad ----------------
+// ---------------- MFMA dgrad v2 (hot path) ----------------
+
+at::Tensor dgrad_weight_decode(const at::Tensor& wp, const at::Tensor& alpha,
+                               int64_t C) {
+  TORCH_CHECK(wp.is_cuda() && wp.dim() == 4 && wp.size(1) == 3 &&
+                  wp.size(2) == 3,
+              "dgrad_weight_decode: packed 3x3 weights [K][3][3][CW]");
+  int K = (int)wp.size(0);
+  TORCH_CHECK(K % 8 == 0 && C % 32 == 0, "dgrad_weight_decode: K%8, C%32");
+  auto wd = at::empty({9, C, K}, wp.options().dtype(at::kBFloat16));
+  auto al = alpha.contiguous();
+  auto wpc = wp.contiguous();
+  bdbnn_dgrad_wdec((const uint32_t*)wpc.data_ptr<int>(),
+                   al.data_ptr<float>(), wd.data_ptr(), (int)C, K,
+                   cur_stream());
+  return wd;
+}
+
+// supported(N,H,W,C,K) mirror of the kernel's launch table
+static bool dgrad2_ok(int H, int W, int C, int K) {
+  if (C % 64 || K % 64 || W > 64) return false;
+  if (W <= 8 && H > 8) return false;
+  return true;
+}
+
+at::Tensor conv_dgrad2(const at::Tensor& g, const at::Tensor& wd,
+                       const at::Tensor& mp, int64_t C) {
+  TORCH_CHECK(g.is_cuda() && g.dim() == 4 &&
+                  g.scalar_type() == at::kBFloat16 &&
+                  g.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "conv_dgrad2: bf16 channels_last grad");
+  int N = (int)g.size(0), K = (int)g.size(1);
+  int H = (int)g.size(2), W = (int)g.size(3);
+  TORCH_CHECK(wd.dim() == 3 && wd.size(0) == 9 && wd.size(1) == C &&
+                  wd.size(2) == K && wd.scalar_type() == at::kBFloat16,
+              "conv_dgrad2: decoded weights [9][C][K] bf16");
+  TORCH_CHECK(mp.dim() == 2 && mp.size(1) == C / 32,
+              "conv_dgrad2: mask bitplane [P][C/32]");
+  TORCH_CHECK(dgrad2_ok(H, W, (int)C, K), "conv_dgrad2: unsupported shape");
+  auto dx = at::empty({N, C, H, W}, g.options(),
+                      at::MemoryFormat::ChannelsLast);
+  int rc = bdbnn_conv_dgrad2(g.data_ptr(), wd.data_ptr(),
+                             (const uint32_t*)mp.data_ptr<int>(),
+                             dx.data_ptr(), N, H, W, (int)C, K,
+                             cur_stream());
+  TORCH_CHECK(rc == 0, "conv_dgrad2: launch rejected the shape");
+  return dx;
+}
+
+// ---------------- experimental MFMA dgrad (v1, kept for A/B) -------------
 
 at::Tensor conv_dgrad(const at::Tensor& g, const at::Tensor& wp,
                       const at::Tensor& alpha, int64_t C) {
@@ -786,6 +839,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused BN(+add)(+act) training forward");
   m.def("bn_act_bwd", &bn_act_bwd, "fused BN(+add)(+act) backward");
   m.def("bn_act_eval", &bn_act_eval, "fused BN(+add)(+act) eval forward");
+  m.def("conv_dgrad2", &conv_dgrad2,
+        "MFMA bf16 dgrad v2: halo-staged 9-tap implicit GEMM with fused "
+        "clip-STE mask (3x3/s1/p1)");
+  m.def("dgrad_weight_decode", &dgrad_weight_decode,
+        "packed bits -> mirrored transposed +-alpha bf16 [9][C][K]");
+  m.def("dgrad2_supported",
+        [](int64_t H, int64_t W, int64_t C, int64_t K) {
+          return dgrad2_ok((int)H, (int)W, (int)C, (int)K);
+        },
+        "shape support predicate for conv_dgrad2");
   m.def("conv_dgrad", &conv_dgrad,
         "EXPERIMENTAL MFMA bf16 dgrad (3x3/s1/p1, packed weights)");
   m.def("conv_wgrad", &conv_wgrad,

@@ -1,0 +1,318 @@
+// MFMA bf16 backward for the 3x3/stride-1/pad-1 binary convolutions —
+// the training hot path's dgrad (v2), replacing both MIOpen igemm and
+// the separate decode/mask passes (SURVEY.md hard-part #1).
+//
+//   dx[n,y,x,c] = mask(n,y,x,c) * sum_{dy,dx,k} g[n, y+dy-1, x+dx-1, k]
+//                                  * Wd[t=(dy,dx)][c][k]
+//
+// where Wd[t][c][k] = alpha_k * sign(w[k,c,2-dy,2-dx]) is the mirrored,
+// transposed, pre-decoded weight (tiny tensor, produced once per
+// backward by dgrad_weight_decode below), and mask is the clip-STE
+// bitplane packed by the forward (csrc/pack.hip) — applied in the
+// epilogue, so the old mask_mul_packed pass over the full dx tensor
+// disappears.
+//
+// Why v2 is fast where v1 (csrc/conv_dgrad.hip) was 4-6x behind MIOpen:
+//   * HALO staging: the block's g tile is staged ONCE per 64-channel
+//     chunk as a 2D zero-padded halo band, and all 9 taps read shifted
+//     windows of it from LDS — v1 re-staged the same g rows 9x.
+//   * 576-deep K per halo stage (9 taps x 64 ch) with ONE barrier per
+//     tap (write-late double buffering, the XNOR-forward structure) —
+//     v1 ran 2 barriers per 16-deep K step.
+//   * XOR-swizzled LDS (guide T2) — conflict-free ds_read_b128.
+//   * raw __builtin_amdgcn_mfma_f32_32x32x16_bf16 with 256x64 block
+//     tile, 8 waves as 4(M)x2(N).
+//
+// Geometry: the 256 M-rows of a block are image-row BANDS in padded
+// x-coordinates (Wp = W rounded up to 8/16/32/64); pad taps read
+// explicit zeros in the halo, dummy columns (x >= W) produce rows that
+// are simply not stored.  One template instantiation per Wp class.
+//
+// Constraints (python falls back to the aten/MIOpen path otherwise):
+//   KH = KW = 3, stride = 1, pad = 1, C % 64 == 0, K % 64 == 0,
+//   W <= 64 (and H <= 8 when W <= 8), g bf16 NHWC, dx bf16 NHWC.
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+#define DG2_BM 256       // M rows (padded pixels) per block
+#define DG2_BN 64        // dx channels per block
+#define DG2_BK 64        // g channels per K-chunk
+
+struct Dgrad2Params {
+  int N, H, W, C, K;
+  int bands_per_image;   // ceil(H / RB)
+  int total_slots;       // N * bands_per_image
+  int CW;                // C/32 (mask words per pixel)
+};
+
+// One template instantiation per padded-width class.
+//   WP: padded row width; RB: image rows per band; GB: bands per block.
+//   GB*(RB)*WP == 256.
+template <int WP, int RB, int GB>
+__global__ __launch_bounds__(512, 2) void conv_dgrad2_kernel(
+    const __bf16* __restrict__ g, const __bf16* __restrict__ wd,
+    const uint32_t* __restrict__ mp, __bf16* __restrict__ dx,
+    Dgrad2Params p, int grid_m) {
+  constexpr int WH = WP + 2;                  // halo row width
+  constexpr int NHE = GB * (RB + 2) * WH;     // halo entries (128 B each)
+  constexpr int PIECES = NHE * 8;             // 16-B staging pieces
+  constexpr int PPT = (PIECES + 511) / 512;   // pieces per thread
+
+  // XCD-aware bijective remap (8 XCDs with private L2s): contiguous
+  // spatial tiles per XCD so neighbouring bands share L2 lines.
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, idx = wg / 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int m_blk = wg % grid_m;
+  const int c_blk = wg / grid_m;
+  const int c0 = c_blk * DG2_BN;
+  const int tid = threadIdx.x;
+
+  __shared__ __align__(16) __bf16 halo[2][NHE * DG2_BK];
+  __shared__ __align__(16) __bf16 blds[2][DG2_BN * DG2_BK];
+  __shared__ int he_base[NHE];     // g element offset of halo entry, or -1
+  __shared__ int tab_he[DG2_BM];   // he00 of m-row (tap dy=dx=0)
+  __shared__ int tab_out[DG2_BM];  // output pixel index, or -1
+  __shared__ unsigned char tab_x7[DG2_BM];  // x & 7 (swizzle key base)
+
+  // ---- per-block tables ----
+  const int slot0 = m_blk * GB;
+  for (int he = tid; he < NHE; he += 512) {
+    int band = he / ((RB + 2) * WH);
+    int rem = he - band * ((RB + 2) * WH);
+    int hr = rem / WH, hx = rem - hr * WH;
+    int slot = slot0 + band;
+    int base = -1;
+    if (slot < p.total_slots) {
+      int n = slot / p.bands_per_image;
+      int r0 = (slot - n * p.bands_per_image) * RB;
+      int y = r0 + hr - 1, x = hx - 1;
+      if (y >= 0 && y < p.H && x >= 0 && x < p.W)
+        base = (n * p.H + y) * p.W + x;   // pixel index (x K at use)
+    }
+    he_base[he] = base;
+  }
+  for (int m = tid; m < DG2_BM; m += 512) {
+    int band = m / (RB * WP);
+    int rem = m - band * (RB * WP);
+    int rloc = rem / WP, x = rem - rloc * WP;
+    int slot = slot0 + band;
+    tab_he[m] = band * (RB + 2) * WH + rloc * WH + x;
+    tab_x7[m] = (unsigned char)(x & 7);
+    int out = -1;
+    if (slot < p.total_slots) {
+      int n = slot / p.bands_per_image;
+      int y = (slot - n * p.bands_per_image) * RB + rloc;
+      if (y < p.H && x < p.W) out = (n * p.H + y) * p.W + x;
+    }
+    tab_out[m] = out;
+  }
+  __syncthreads();
+
+  // ---- wave decomposition: 8 waves as 4(M) x 2(N) ----
+  const int wid = tid >> 6;
+  const int lane = tid & 63;
+  const int wm0 = (wid >> 1) * 64;       // wave's M offset (64 rows)
+  const int wc = (wid & 1) * 32;         // wave's c offset (32 cols)
+  const int lrow = lane & 31;
+  const int lk8 = lane >> 5;             // which 8-element k-half
+
+  // per-lane A addressing for its two 32-row M-fragments
+  const int mA0 = wm0 + lrow, mA1 = wm0 + 32 + lrow;
+  const int heA0 = tab_he[mA0], heA1 = tab_he[mA1];
+  const int x7A0 = tab_x7[mA0], x7A1 = tab_x7[mA1];
+  // B addressing: lane's c row
+  const int cB = wc + lrow;
+
+  f32x16 acc0 = {0.f}, acc1 = {0.f};
+#pragma unroll
+  for (int i = 0; i < 16; ++i) { acc0[i] = 0.f; acc1[i] = 0.f; }
+
+  // ---- staging (registers -> LDS, write-late) ----
+  uint4 hreg[PPT];
+  uint4 breg;
+
+#define HALO_LOAD(k0)                                                     \
+  {                                                                       \
+    _Pragma("unroll")                                                     \
+    for (int it = 0; it < PPT; ++it) {                                    \
+      int i = tid + it * 512;                                             \
+      uint4 v{0, 0, 0, 0};                                                \
+      if (i < PIECES) {                                                   \
+        int he = i >> 3, k8 = i & 7;                                      \
+        int base = he_base[he];                                           \
+        if (base >= 0)                                                    \
+          v = *(const uint4*)(g + (int64_t)base * p.K + (k0) + k8 * 8);   \
+      }                                                                   \
+      hreg[it] = v;                                                       \
+    }                                                                     \
+  }
+
+#define HALO_WRITE(buf)                                                   \
+  {                                                                       \
+    _Pragma("unroll")                                                     \
+    for (int it = 0; it < PPT; ++it) {                                    \
+      int i = tid + it * 512;                                             \
+      if (i < PIECES) {                                                   \
+        int he = i >> 3, k8 = i & 7;                                      \
+        int rem = he - (he / WH) * WH;  /* hx */                          \
+        *(uint4*)&halo[buf][he * DG2_BK + ((k8 ^ (rem & 7)) << 3)] =      \
+            hreg[it];                                                     \
+      }                                                                   \
+    }                                                                     \
+  }
+
+  // B tile for tap t: wd[t][c0+c][k0..k0+63]; thread loads one 16-B piece
+  const int b_c = tid >> 3;        // 0..63
+  const int b_k8 = tid & 7;        // 0..7
+#define B_LOAD(t, k0)                                                     \
+  breg = *(const uint4*)(wd + (((t) * p.C + c0 + b_c) * (int64_t)p.K) +   \
+                         (k0) + b_k8 * 8);
+#define B_WRITE(buf)                                                      \
+  *(uint4*)&blds[buf][b_c * DG2_BK + ((b_k8 ^ (b_c & 7)) << 3)] = breg;
+
+  HALO_LOAD(0);
+  HALO_WRITE(0);
+  B_LOAD(0, 0);
+  B_WRITE(0);
+  __syncthreads();
+
+  int hb = 0, bb = 0;
+  const int n_k0 = p.K / DG2_BK;
+  for (int ki = 0; ki < n_k0; ++ki) {
+    const int k0 = ki * DG2_BK;
+    const bool more = ki + 1 < n_k0;
+    if (more) HALO_LOAD(k0 + DG2_BK);   // issue early, write at tap 8
+#pragma unroll
+    for (int t = 0; t < 9; ++t) {
+      const int dy = t / 3, dxt = t - dy * 3;
+      // prefetch next B tile (next tap, or tap 0 of the next chunk)
+      if (t < 8) {
+        B_LOAD(t + 1, k0);
+      } else if (more) {
+        B_LOAD(0, k0 + DG2_BK);
+      }
+      // this tap's A base addresses (element offsets into halo[hb])
+      const int heT0 = (heA0 + dy * WH + dxt) * DG2_BK;
+      const int heT1 = (heA1 + dy * WH + dxt) * DG2_BK;
+      const int keyA0 = ((x7A0 + dxt) & 7) ^ lk8;
+      const int keyA1 = ((x7A1 + dxt) & 7) ^ lk8;
+      const int keyB = (cB & 7) ^ lk8;
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        // kk8 index of this 16-deep step's k-half = kk*2 ^ swizzle key
+        bf16x8 a0 = *(const bf16x8*)&halo[hb][heT0 + (((kk << 1) ^ keyA0) << 3)];
+        bf16x8 a1 = *(const bf16x8*)&halo[hb][heT1 + (((kk << 1) ^ keyA1) << 3)];
+        bf16x8 b = *(const bf16x8*)&blds[bb][cB * DG2_BK + (((kk << 1) ^ keyB) << 3)];
+        acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b, acc0, 0, 0, 0);
+        acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b, acc1, 0, 0, 0);
+      }
+      if (t == 8 && more) HALO_WRITE(hb ^ 1);
+      if (t < 8 || more) {
+        B_WRITE(bb ^ 1);
+        __syncthreads();
+        bb ^= 1;
+      }
+    }
+    hb ^= 1;
+  }
+
+  // ---- epilogue: clip-STE mask from the packed bitplane, bf16 store ----
+  // C/D layout (guide section 3): col = lane&31, row = (reg&3) + 8*(reg>>2)
+  // + 4*(lane>>5).
+  const int ccol = c0 + wc + lrow;
+  const int cw_word = ccol >> 5;              // uniform per 32-lane half
+  const int cbit = ccol & 31;
+#pragma unroll
+  for (int half = 0; half < 2; ++half) {
+    const f32x16& acc = half ? acc1 : acc0;
+    const int mbase = wm0 + half * 32 + 4 * lk8;
+#pragma unroll
+    for (int reg = 0; reg < 16; ++reg) {
+      int m = mbase + (reg & 3) + 8 * (reg >> 2);
+      int pix = tab_out[m];
+      if (pix < 0) continue;
+      uint32_t word = mp[(int64_t)pix * p.CW + cw_word];
+      float v = (word >> cbit) & 1 ? acc[reg] : 0.f;
+      uint16_t h = f32_to_bf16(v);
+      *(uint16_t*)(dx + (int64_t)pix * p.C + ccol) = h;
+    }
+  }
+}
+
+// ---------------- host launcher ----------------
+
+extern "C" int bdbnn_conv_dgrad2(const void* g, const void* wd,
+                                 const uint32_t* mp, void* dx, int N, int H,
+                                 int W, int C, int K, hipStream_t stream) {
+  if (C % 64 || K % 64 || W > 64) return -1;
+  Dgrad2Params p;
+  p.N = N; p.H = H; p.W = W; p.C = C; p.K = K;
+  p.CW = C / 32;
+  int grid_m;
+#define LAUNCH(WPV, RBV, GBV)                                             \
+  {                                                                       \
+    p.bands_per_image = (H + (RBV)-1) / (RBV);                            \
+    p.total_slots = N * p.bands_per_image;                                \
+    grid_m = (p.total_slots + (GBV)-1) / (GBV);                           \
+    dim3 grid(grid_m * (C / DG2_BN));                                     \
+    conv_dgrad2_kernel<WPV, RBV, GBV><<<grid, 512, 0, stream>>>(          \
+        (const __bf16*)g, (const __bf16*)wd, mp, (__bf16*)dx, p, grid_m); \
+    return 0;                                                             \
+  }
+  if (W <= 8) {
+    if (H > 8) return -1;
+    LAUNCH(8, 8, 4);
+  } else if (W <= 16) {
+    LAUNCH(16, 16, 1);
+  } else if (W <= 32) {
+    LAUNCH(32, 8, 1);
+  } else {
+    LAUNCH(64, 4, 1);
+  }
+#undef LAUNCH
+}
+
+// ---------------- mirrored/transposed weight decode ----------------
+// wp: uint32 [K][9][CW] inverted bits (csrc/pack.hip: bit 1 <=> w < 0),
+// alpha: fp32 [K]  ->  wd bf16 [9][C][K], wd[t][c][k] = +-alpha_k with
+// the tap MIRRORED (t reads w[k,c,2-dy,2-dx]).
+__global__ void dgrad_wdec_kernel(const uint32_t* __restrict__ wp,
+                                  const float* __restrict__ alpha,
+                                  __bf16* __restrict__ wd, int C, int K,
+                                  int CW) {
+  // one thread per 8 outputs along k at fixed (t, c)
+  int64_t total = (int64_t)9 * C * (K / 8);
+  GRID_STRIDE(i, total) {
+    int k8 = int(i % (K / 8));
+    int64_t rem = i / (K / 8);
+    int c = int(rem % C);
+    int t = int(rem / C);
+    int tm = 8 - t;                       // mirrored tap
+    int cw = c >> 5, cb = c & 31;
+    __bf16 v[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int k = k8 * 8 + j;
+      uint32_t bits = wp[((int64_t)k * 9 + tm) * CW + cw];
+      float al = alpha[k];
+      v[j] = (__bf16)((bits >> cb) & 1 ? -al : al);
+    }
+    *(uint4*)&wd[((int64_t)t * C + c) * K + k8 * 8] = *(uint4*)v;
+  }
+}
+
+extern "C" void bdbnn_dgrad_wdec(const uint32_t* wp, const float* alpha,
+                                 void* wd, int C, int K,
+                                 hipStream_t stream) {
+  int64_t total = (int64_t)9 * C * (K / 8);
+  int blocks = (int)bd_min<int64_t>((total + 255) / 256, 4096);
+  dgrad_wdec_kernel<<<blocks, 256, 0, stream>>>(wp, alpha, (__bf16*)wd, C,
+                                                K, C / 32);
+}

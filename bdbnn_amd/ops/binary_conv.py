@@ -32,6 +32,9 @@ from .binarize import binsign, weight_scale
 from .. import _C
 
 _CONV_STATS = os.environ.get("BDBNN_CONV_STATS", "0") == "1"
+# hand-written MFMA backward kernels (default ON; BDBNN_MFMA_BWD=0 falls
+# back to MIOpen igemm on decoded operands for A/B comparison)
+_MFMA_BWD = os.environ.get("BDBNN_MFMA_BWD", "1") != "0"
 
 
 def _act_grad_mask(x: torch.Tensor, mode: str, t, k) -> torch.Tensor:
@@ -116,14 +119,33 @@ class BinaryConvFunction(torch.autograd.Function):
             w, xp, mp, wp, alpha = ctx.saved_tensors
             nat = _C.native_required()
             bf16 = g.dtype == torch.bfloat16
-            xb = nat.decode_packed(xp, ctx.in_channels, bf16)
-            wb = nat.weight_decode(wp, alpha, ctx.in_channels, bf16)
+            C = ctx.in_channels
+            K, kh = w.shape[0], w.shape[2]
             g = g.contiguous(memory_format=torch.channels_last)
+            H, W = g.shape[2], g.shape[3]
+            # hand-written MFMA dgrad (halo implicit GEMM, clip-STE mask
+            # fused into the epilogue) — the default on the hot path
+            use2 = (_MFMA_BWD and bf16 and ctx.x_dtype == torch.bfloat16
+                    and stride == 1 and padding == 1 and kh == 3
+                    and nat.dgrad2_supported(H, W, C, K))
+            if use2:
+                wd = nat.dgrad_weight_decode(wp, alpha, C)
+                dx = nat.conv_dgrad2(g, wd, mp, C)
+                xb = nat.decode_packed(xp, C, bf16)
+                wb = nat.weight_decode(wp, alpha, C, bf16)
+                dwb = torch.ops.aten.convolution_backward(
+                    g, xb, wb, None, [stride, stride], [padding, padding],
+                    [1, 1], False, [0, 0], 1, [False, True, False])[1]
+                dw = nat.ste_mask_mul(dwb, w, 0, 0.0, 0.0)
+                return (dx, dw.to(w.dtype), None, None, None, None, None,
+                        None)
+            xb = nat.decode_packed(xp, C, bf16)
+            wb = nat.weight_decode(wp, alpha, C, bf16)
             dxb, dwb = torch.ops.aten.convolution_backward(
                 g, xb, wb, None,
                 [stride, stride], [padding, padding], [1, 1], False, [0, 0],
                 1, [True, True, False])[:2]
-            dx = nat.mask_mul_packed(dxb, mp, ctx.in_channels,
+            dx = nat.mask_mul_packed(dxb, mp, C,
                                      ctx.x_dtype == torch.bfloat16)
             dw = nat.ste_mask_mul(dwb, w, 0, 0.0, 0.0)
             return dx, dw.to(w.dtype), None, None, None, None, None, None

@@ -26,6 +26,7 @@ sources = [os.path.join(CSRC, f) for f in (
     "pool.hip",
     "loss.hip",
     "conv_dgrad.hip",
+    "conv_bwd.hip",
     "conv_wgrad.hip",
     "xnor_conv.hip",
     "kurtosis.hip",

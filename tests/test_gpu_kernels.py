@@ -622,3 +622,80 @@ def test_vgg_small_gpu_step():
     loss.backward()
     opt.step()
     assert torch.isfinite(loss)
+
+
+# ---------------- MFMA dgrad v2 (hot path) ----------------
+
+def _dgrad2_ref(g, w, mask):
+    """fp32 oracle: conv_transpose with alpha*sign(w) + clip-STE mask."""
+    wb = (weight_scale(w) * binsign(w)).to(torch.bfloat16)
+    ref = torch.nn.functional.conv_transpose2d(
+        g.float(), wb.float(), None, stride=1, padding=1)
+    return ref * mask.float()
+
+
+@pytest.mark.parametrize("N,C,H,K", [
+    (2, 64, 56, 64),     # Wp=64 class (stage 1)
+    (2, 128, 28, 128),   # Wp=32
+    (2, 256, 14, 256),   # Wp=16
+    (4, 512, 7, 512),    # Wp=8 (multi-image blocks)
+    (1, 64, 9, 64),      # Wp=16 with W<Wp (dummy columns + partial band)
+    (3, 64, 30, 64),     # Wp=32, H%Rb != 0 (dummy band rows), odd N
+])
+def test_conv_dgrad2_matches_reference(N, C, H, K):
+    torch.manual_seed(31)
+    nat = _nat()
+    g = _cl(torch.randn(N, K, H, H, device="cuda", dtype=torch.bfloat16))
+    w = torch.randn(K, C, 3, 3, device="cuda")
+    x = torch.randn(N, C, H, H, device="cuda")
+    wp, alpha, _ = nat.weight_pack(w)
+    _, mp = nat.sign_mask_pack_nhwc(_cl(x))
+    assert nat.dgrad2_supported(H, H, C, K)
+    wd = nat.dgrad_weight_decode(wp, alpha, C)
+    dx = nat.conv_dgrad2(g, wd, mp, C)
+    mask = (x.abs() <= 1).to(torch.float32)
+    ref = _dgrad2_ref(g, w, mask)
+    assert dx.shape == ref.shape
+    err = (dx.float() - _cl(ref)).abs().max().item()
+    assert torch.allclose(dx.float(), _cl(ref), atol=0.5, rtol=2e-2), err
+
+
+def test_dgrad_weight_decode_values():
+    torch.manual_seed(32)
+    nat = _nat()
+    K, C = 64, 64
+    w = torch.randn(K, C, 3, 3, device="cuda")
+    wp, alpha, _ = nat.weight_pack(w)
+    wd = nat.dgrad_weight_decode(wp, alpha, C)   # [9][C][K]
+    want = (weight_scale(w) * binsign(w)).to(torch.bfloat16)  # [K][C][3][3]
+    for t in range(9):
+        dy, dx_ = t // 3, t % 3
+        ref = want[:, :, 2 - dy, 2 - dx_].transpose(0, 1)  # [C][K] mirrored
+        assert torch.equal(wd[t], ref)
+
+
+def test_packed_backward_uses_mfma_dgrad():
+    """Full BinaryConvFunction backward: MFMA-on vs MIOpen fallback agree."""
+    import bdbnn_amd.ops.binary_conv as bc
+    torch.manual_seed(33)
+    x = _cl(torch.randn(2, 64, 14, 14, device="cuda",
+                        dtype=torch.bfloat16))
+    w = torch.randn(64, 64, 3, 3, device="cuda")
+
+    def run(flag):
+        old = bc._MFMA_BWD
+        bc._MFMA_BWD = flag
+        try:
+            xl = x.clone().requires_grad_(True)
+            wl = w.clone().requires_grad_(True)
+            out, _, _ = BinaryConvFunction.apply(xl, wl, 1, 1, "ste", None,
+                                                 None, False)
+            out.float().pow(2).sum().backward()
+            return xl.grad.float(), wl.grad.float()
+        finally:
+            bc._MFMA_BWD = old
+    dx2, dw2 = run(True)
+    dx1, dw1 = run(False)
+    assert torch.allclose(dx2, dx1, atol=1e-2, rtol=1e-2), \
+        (dx2 - dx1).abs().max().item()
+    assert torch.allclose(dw2, dw1, atol=1e-2, rtol=1e-2)
