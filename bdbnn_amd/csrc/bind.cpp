@@ -58,6 +58,12 @@ int bdbnn_conv_dgrad2(const void*, const void*, const uint32_t*, void*,
                       int, int, int, int, int, hipStream_t);
 void bdbnn_dgrad_wdec(const uint32_t*, const float*, void*, int, int,
                       hipStream_t);
+int bdbnn_conv_wgrad2(const void*, const uint64_t*, float*, int, int, int,
+                      int, int, hipStream_t);
+void bdbnn_repack_cplane(const uint32_t*, uint64_t*, int, int, int, int,
+                         hipStream_t);
+void bdbnn_wgrad_finish(const float*, const float*, float*, int, int,
+                        hipStream_t);
 void bdbnn_conv_wgrad(const void*, const uint32_t*, float*, int, int, int,
                       int, int, int, hipStream_t);
 void bdbnn_prelu_fwd(const void*, const float*, void*, int64_t, int, bool,
@@ -647,8 +653,9 @@ at::Tensor conv_dgrad2(const at::Tensor& g, const at::Tensor& wd,
   TORCH_CHECK(wd.dim() == 3 && wd.size(0) == 9 && wd.size(1) == C &&
                   wd.size(2) == K && wd.scalar_type() == at::kBFloat16,
               "conv_dgrad2: decoded weights [9][C][K] bf16");
-  TORCH_CHECK(mp.dim() == 2 && mp.size(1) == C / 32,
-              "conv_dgrad2: mask bitplane [P][C/32]");
+  TORCH_CHECK(mp.size(-1) == C / 32 &&
+                  mp.numel() == (int64_t)N * H * W * (C / 32),
+              "conv_dgrad2: mask bitplane [...P...][C/32]");
   TORCH_CHECK(dgrad2_ok(H, W, (int)C, K), "conv_dgrad2: unsupported shape");
   auto dx = at::empty({N, C, H, W}, g.options(),
                       at::MemoryFormat::ChannelsLast);
@@ -658,6 +665,59 @@ at::Tensor conv_dgrad2(const at::Tensor& g, const at::Tensor& wd,
                              cur_stream());
   TORCH_CHECK(rc == 0, "conv_dgrad2: launch rejected the shape");
   return dx;
+}
+
+// ---------------- MFMA wgrad v2 (hot path) ----------------
+
+at::Tensor repack_cplane(const at::Tensor& xp, int64_t C, int64_t W) {
+  TORCH_CHECK(xp.is_cuda() && xp.dim() == 4 && xp.size(3) == (C + 31) / 32,
+              "repack_cplane: packed activations [N][H][W][CW]");
+  TORCH_CHECK(xp.size(2) == W, "repack_cplane: W mismatch");
+  TORCH_CHECK(W <= 64, "repack_cplane: W <= 64");
+  int NH = (int)(xp.size(0) * xp.size(1));
+  auto xcp = at::empty({C, NH}, xp.options().dtype(at::kLong));
+  bdbnn_repack_cplane((const uint32_t*)xp.data_ptr<int>(),
+                      (uint64_t*)xcp.data_ptr<int64_t>(), NH, (int)W,
+                      (int)C, (int)xp.size(3), cur_stream());
+  return xcp;
+}
+
+at::Tensor conv_wgrad2(const at::Tensor& g, const at::Tensor& xcp,
+                       int64_t C) {
+  TORCH_CHECK(g.is_cuda() && g.dim() == 4 &&
+                  g.scalar_type() == at::kBFloat16 &&
+                  g.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "conv_wgrad2: bf16 channels_last grad");
+  int N = (int)g.size(0), K = (int)g.size(1);
+  int H = (int)g.size(2), W = (int)g.size(3);
+  TORCH_CHECK(xcp.dim() == 2 && xcp.size(0) == C &&
+                  xcp.size(1) == (int64_t)N * H,
+              "conv_wgrad2: c-plane bitplanes [C][N*H]");
+  TORCH_CHECK(C % 64 == 0 && K % 64 == 0 && W <= 64,
+              "conv_wgrad2: unsupported shape");
+  auto dwT = at::zeros({9, C, K}, g.options().dtype(at::kFloat));
+  int rc = bdbnn_conv_wgrad2(g.data_ptr(),
+                             (const uint64_t*)xcp.data_ptr<int64_t>(),
+                             dwT.data_ptr<float>(), N, H, W, (int)C, K,
+                             cur_stream());
+  TORCH_CHECK(rc == 0, "conv_wgrad2: launch rejected the shape");
+  return dwT;
+}
+
+at::Tensor wgrad_finish(const at::Tensor& dwT, const at::Tensor& w) {
+  TORCH_CHECK(w.is_cuda() && w.dim() == 4 && w.size(2) == 3 &&
+                  w.size(3) == 3 && w.scalar_type() == at::kFloat,
+              "wgrad_finish: fp32 [K][C][3][3] latent weights");
+  int K = (int)w.size(0), C = (int)w.size(1);
+  TORCH_CHECK(dwT.dim() == 3 && dwT.size(0) == 9 && dwT.size(1) == C &&
+                  dwT.size(2) == K && dwT.scalar_type() == at::kFloat,
+              "wgrad_finish: dwT [9][C][K] fp32");
+  auto wc = w.contiguous();
+  auto dw = at::empty_like(wc);
+  bdbnn_wgrad_finish(dwT.contiguous().data_ptr<float>(),
+                     wc.data_ptr<float>(), dw.data_ptr<float>(), C, K,
+                     cur_stream());
+  return dw;
 }
 
 // ---------------- experimental MFMA dgrad (v1, kept for A/B) -------------
@@ -849,6 +909,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           return dgrad2_ok((int)H, (int)W, (int)C, (int)K);
         },
         "shape support predicate for conv_dgrad2");
+  m.def("conv_wgrad2", &conv_wgrad2,
+        "MFMA bf16 wgrad v2: all-9-tap block GEMM from sign BITS "
+        "(3x3/s1/p1) -> fp32 [9][C][K]");
+  m.def("repack_cplane", &repack_cplane,
+        "activation sign bits -> padded per-channel u64 row planes");
+  m.def("wgrad_finish", &wgrad_finish,
+        "dwT transpose to [K][C][3][3] + |w|<=1 STE mask, one pass");
   m.def("conv_dgrad", &conv_dgrad,
         "EXPERIMENTAL MFMA bf16 dgrad (3x3/s1/p1, packed weights)");
   m.def("conv_wgrad", &conv_wgrad,

@@ -316,3 +316,321 @@ extern "C" void bdbnn_dgrad_wdec(const uint32_t* wp, const float* alpha,
   dgrad_wdec_kernel<<<blocks, 256, 0, stream>>>(wp, alpha, (__bf16*)wd, C,
                                                 K, C / 32);
 }
+
+// ======================= wgrad v2 =======================
+//
+//   dwT[t][c][k] = sum_m xb[pix(m) + Delta_t][c] * g[pix(m)][k]
+//
+// computed TRANSPOSED (D[c][k] per tap) so both MFMA operands read
+// contiguous along the reduction (m) axis:
+//   A[c][m] = +-1 decoded from the PADDED C-PLANE bitplanes (xcp, built
+//             by repack_cplane below) via a 256-entry byte->8xbf16 LDS
+//             LUT, into three kw-SHIFTED copies (X0/X1/X2) so every tap
+//             read is aligned;  pad taps are explicit zeros.
+//   B[m][k] = g staged TRANSPOSED [k][m] in LDS.
+// All 9 taps accumulate in one block pass over the chunk (g and xb are
+// read from HBM exactly once; xb moves as BITS - 32x less traffic than
+// the decoded tensor the MIOpen path reads).  fp32 atomicAdd combine
+// into dwT[9][C][K]; wgrad_finish transposes to [K][C][3][3] and
+// applies the |w|<=1 STE mask in the same pass.
+//
+// Constraints: 3x3/s1/p1, C % 64 == 0, K % 64 == 0, W <= 64, g bf16
+// channels_last.
+
+#define WG2_CHUNK 128       // m entries per chunk
+#define WG2_BC 64           // c tile
+#define WG2_BK 64           // k tile
+
+struct Wgrad2Params {
+  int N, H, W, C, K;
+  int chunks_per_image;     // ceil(H / rc)
+  int total_chunks;         // N * chunks_per_image
+  int split;                // chunk stride (== blocks per (c,k) tile)
+  int nh_rows;              // N * H  (xcp row count)
+};
+
+template <int WP>
+__global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
+    const __bf16* __restrict__ g, const uint64_t* __restrict__ xcp,
+    float* __restrict__ dwT, Wgrad2Params p, int grid_ck) {
+  constexpr int RC = WG2_CHUNK / WP;      // image rows per chunk
+  constexpr int XROWS = RC + 2;           // halo rows decoded per chunk
+  constexpr int XSTRIDE = XROWS * WP + 8; // +16 B pad: bank-conflict-free
+  constexpr int GSTRIDE = WG2_CHUNK + 8;  // gT row stride (elements)
+  constexpr int LGWP = WP == 8 ? 3 : WP == 16 ? 4 : WP == 32 ? 5 : 6;
+
+  const int tid = threadIdx.x;
+  int wg = blockIdx.x;
+  const int ck = wg % grid_ck;            // (c,k) tile
+  const int split_id = wg / grid_ck;
+  const int c_blk = ck % (p.C / WG2_BC);
+  const int k_blk = ck / (p.C / WG2_BC);
+  const int c0 = c_blk * WG2_BC, k0 = k_blk * WG2_BK;
+
+  // X: [3 copies][64 c] rows of XROWS*WP (+pad) elements
+  __shared__ __align__(16) __bf16 X[3 * WG2_BC * XSTRIDE];
+  __shared__ __align__(16) __bf16 gT[2][WG2_BK * GSTRIDE];
+  __shared__ __align__(16) __bf16 lut[256][8];
+  __shared__ float red[4][32][32];        // m-split combine scratch
+
+  // ---- byte -> 8 x (+-1) LUT (bit 1 <=> x >= 0 <=> +1) ----
+  for (int b = tid; b < 256; b += 512) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      lut[b][j] = (__bf16)(((b >> j) & 1) ? 1.f : -1.f);
+  }
+
+  // ---- wave decomposition: quad (c-half, k-half) x m-split ----
+  const int wid = tid >> 6, lane = tid & 63;
+  const int qc = (wid >> 1) & 1, qk = wid & 1, ms_grp = wid >> 2;
+  const int lrow = lane & 31, lhalf = lane >> 5;
+
+  f32x16 acc[9];
+#pragma unroll
+  for (int t = 0; t < 9; ++t)
+#pragma unroll
+    for (int i = 0; i < 16; ++i) acc[t][i] = 0.f;
+
+  // staging assignments (fixed per thread)
+  //   gT: thread loads 16 g elements (two 16-B pieces along k at fixed m)
+  //       and scatter-writes them transposed.
+  const int sg_m = tid >> 2;              // 0..127
+  const int sg_k16 = (tid & 3) * 16;      // k offset of its 16 elements
+  // X decode: targets (copy dxs, c, row, 8-x block), grid-strided.
+  constexpr int XTGT = 3 * WG2_BC * XROWS * (WP / 8);
+
+  const int n_chunks = (p.total_chunks - split_id + p.split - 1) / p.split;
+  int ch = split_id;
+
+  uint4 greg[2];
+
+#define G_LOAD(chunk)                                                     \
+  {                                                                       \
+    int n = (chunk) / p.chunks_per_image;                                 \
+    int y0 = ((chunk) - n * p.chunks_per_image) * RC;                     \
+    int y = y0 + (sg_m >> LGWP);                                          \
+    int x = sg_m & (WP - 1);                                              \
+    uint4 z{0, 0, 0, 0};                                                  \
+    greg[0] = z; greg[1] = z;                                             \
+    if (x < p.W && y < p.H) {                                             \
+      const __bf16* src =                                                 \
+          g + ((int64_t)(n * p.H + y) * p.W + x) * p.K + k0 + sg_k16;     \
+      greg[0] = *(const uint4*)src;                                       \
+      greg[1] = *(const uint4*)(src + 8);                                 \
+    }                                                                     \
+  }
+
+#define G_WRITE(buf)                                                      \
+  {                                                                       \
+    _Pragma("unroll")                                                     \
+    for (int j = 0; j < 16; ++j) {                                        \
+      __bf16 v = ((const __bf16*)greg)[j];                                \
+      gT[buf][(sg_k16 + j) * GSTRIDE + sg_m] = v;                         \
+    }                                                                     \
+  }
+
+  // ---- X decode for chunk ch (LDS-resident after barrier) ----
+#define X_DECODE(chunk)                                                   \
+  {                                                                       \
+    int n = (chunk) / p.chunks_per_image;                                 \
+    int y0 = ((chunk) - n * p.chunks_per_image) * RC;                     \
+    _Pragma("unroll 2")                                                   \
+    for (int it = 0; it < (XTGT + 511) / 512; ++it) {                     \
+      int i = tid + it * 512;                                             \
+      if (i < XTGT) {                                                     \
+        int xb8 = i % (WP / 8);                                           \
+        int rem = i / (WP / 8);                                           \
+        int row = rem % XROWS;                                            \
+        rem /= XROWS;                                                     \
+        int c = rem % WG2_BC;                                             \
+        int dxs = rem / WG2_BC;                                           \
+        int y = y0 + row - 1;                                             \
+        uint4 vv{0, 0, 0, 0};                                             \
+        if (y >= 0 && y < p.H) {                                          \
+          /* y-pad rows stay ZERO (a 0-bit byte would decode to -1) */    \
+          uint64_t bits =                                                 \
+              xcp[(int64_t)(c0 + c) * p.nh_rows + n * p.H + y];           \
+          /* kw shift: X_dxs[x] = xb[x + dxs - 1] */                      \
+          uint64_t sh = dxs == 0 ? (bits << 1) : (bits >> (dxs - 1));     \
+          unsigned byte = (unsigned)(sh >> (8 * xb8)) & 0xffu;            \
+          vv = *(const uint4*)&lut[byte][0];                              \
+        }                                                                 \
+        *(uint4*)&X[(dxs * WG2_BC + c) * XSTRIDE + row * WP + xb8 * 8] =  \
+            vv;                                                           \
+      }                                                                   \
+    }                                                                     \
+    __syncthreads();                                                      \
+    /* edge zeros: X0[x=0] (left pad) and X2[x=W-1] (right pad) */        \
+    for (int i = tid; i < 2 * WG2_BC * XROWS; i += 512) {                 \
+      int row = i % XROWS;                                                \
+      int rem = i / XROWS;                                                \
+      int c = rem % WG2_BC;                                               \
+      int which = rem / WG2_BC;                                           \
+      if (which == 0)                                                     \
+        X[(0 * WG2_BC + c) * XSTRIDE + row * WP + 0] = (__bf16)0.f;       \
+      else                                                                \
+        X[(2 * WG2_BC + c) * XSTRIDE + row * WP + (p.W - 1)] =            \
+            (__bf16)0.f;                                                  \
+    }                                                                     \
+  }
+
+  if (n_chunks > 0) {
+    G_LOAD(ch);
+    G_WRITE(0);
+  }
+
+  // per-lane fragment bases
+  // A (X): lane row c = qc*32 + lrow; m-run start offset = lhalf*8
+  const int a_c = qc * 32 + lrow;
+  // B (gT): lane row k = qk*32 + lrow
+  const int b_k = qk * 32 + lrow;
+
+  int gb = 0;
+  for (int ci = 0; ci < n_chunks; ++ci) {
+    const bool more = ci + 1 < n_chunks;
+    // decode this chunk's X (also serves as the gT visibility barrier)
+    X_DECODE(ch);
+    if (more) G_LOAD(ch + p.split);   // issue next g early
+    __syncthreads();                  // X + gT[gb] ready for all waves
+
+#pragma unroll
+    for (int msl = 0; msl < 2; ++msl) {
+      const int m16 = ms_grp * 64 + msl * 16;
+      // B-frag: 8 m at fixed k from gT
+      bf16x8 bfrag = *(const bf16x8*)&gT[gb][b_k * GSTRIDE + m16 +
+                                           lhalf * 8];
+      // A-frags per tap from X: row (rl + dy), x-run
+      const int mstart = m16 + lhalf * 8;
+      const int rl = mstart >> LGWP;
+      const int xs = mstart & (WP - 1);
+#pragma unroll
+      for (int t = 0; t < 9; ++t) {
+        const int dy = t / 3, dxs = t - dy * 3;
+        bf16x8 afrag = *(const bf16x8*)&X[(dxs * WG2_BC + a_c) * XSTRIDE +
+                                          (rl + dy) * WP + xs];
+        acc[t] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(afrag, bfrag, acc[t],
+                                                    0, 0, 0);
+      }
+    }
+    __syncthreads();                  // all reads of X/gT[gb] done
+    if (more) G_WRITE(gb ^ 1);
+    gb ^= 1;
+    ch += p.split;
+  }
+
+  // ---- m-split combine + global atomic add ----
+  // C/D layout: col j (k) = lane&31, row (c) = (reg&3)+8*(reg>>2)+4*lhalf
+  const int quad = wid & 3;
+#pragma unroll
+  for (int t = 0; t < 9; ++t) {
+    __syncthreads();
+    if (ms_grp == 1) {
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        int row = (reg & 3) + 8 * (reg >> 2) + 4 * lhalf;
+        red[quad][row][lrow] = acc[t][reg];
+      }
+    }
+    __syncthreads();
+    if (ms_grp == 0) {
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        int row = (reg & 3) + 8 * (reg >> 2) + 4 * lhalf;
+        float v = acc[t][reg] + red[quad][row][lrow];
+        if (v != 0.f)
+          atomicAdd(&dwT[((int64_t)t * p.C + c0 + qc * 32 + row) * p.K +
+                         k0 + qk * 32 + lrow],
+                    v);
+      }
+    }
+  }
+}
+
+extern "C" int bdbnn_conv_wgrad2(const void* g, const uint64_t* xcp,
+                                 float* dwT, int N, int H, int W, int C,
+                                 int K, hipStream_t stream) {
+  if (C % 64 || K % 64 || W > 64) return -1;
+  Wgrad2Params p;
+  p.N = N; p.H = H; p.W = W; p.C = C; p.K = K;
+  p.nh_rows = N * H;
+  int grid_ck = (C / WG2_BC) * (K / WG2_BK);
+  int split = (384 + grid_ck - 1) / grid_ck;
+  p.split = split;
+#define WLAUNCH(WPV)                                                      \
+  {                                                                       \
+    constexpr int rc = WG2_CHUNK / WPV;                                   \
+    p.chunks_per_image = (H + rc - 1) / rc;                               \
+    p.total_chunks = N * p.chunks_per_image;                              \
+    if (p.split > p.total_chunks) p.split = p.total_chunks;               \
+    dim3 grid(grid_ck * p.split);                                         \
+    conv_wgrad2_kernel<WPV><<<grid, 512, 0, stream>>>(                    \
+        (const __bf16*)g, xcp, dwT, p, grid_ck);                          \
+    return 0;                                                             \
+  }
+  if (W <= 8) WLAUNCH(8)
+  else if (W <= 16) WLAUNCH(16)
+  else if (W <= 32) WLAUNCH(32)
+  else WLAUNCH(64)
+#undef WLAUNCH
+}
+
+// ---------------- padded c-plane repack ----------------
+// xp [P][CW] (bit c of word = sign of x[p, 32cw+c])  ->  xcp u64 rows
+// [C][N*H], row = W bits of one image row zero-padded to 64.
+__global__ void repack_cplane_kernel(const uint32_t* __restrict__ xp,
+                                     uint64_t* __restrict__ xcp, int NH,
+                                     int W, int C, int CW) {
+  // one thread per (c-word cw, image row): builds 32 u64 rows bit by bit
+  GRID_STRIDE(i, (int64_t)NH * CW) {
+    int cw = int(i % CW);
+    int64_t row = i / CW;
+    const uint32_t* src = xp + (row * W) * CW + cw;
+    uint64_t acc[32];
+#pragma unroll
+    for (int c = 0; c < 32; ++c) acc[c] = 0;
+    for (int x = 0; x < W; ++x) {
+      uint32_t word = src[(int64_t)x * CW];
+#pragma unroll
+      for (int c = 0; c < 32; ++c)
+        acc[c] |= (uint64_t)((word >> c) & 1) << x;
+    }
+#pragma unroll
+    for (int c = 0; c < 32; ++c) {
+      int cc = cw * 32 + c;
+      if (cc < C) xcp[(int64_t)cc * NH + row] = acc[c];
+    }
+  }
+}
+
+extern "C" void bdbnn_repack_cplane(const uint32_t* xp, uint64_t* xcp,
+                                    int NH, int W, int C, int CW,
+                                    hipStream_t stream) {
+  int64_t total = (int64_t)NH * CW;
+  int blocks = (int)bd_min<int64_t>((total + 255) / 256, 8192);
+  repack_cplane_kernel<<<blocks, 256, 0, stream>>>(xp, xcp, NH, W, C, CW);
+}
+
+// ---------------- wgrad finish: transpose + STE mask ----------------
+// dw[k][c][t] = dwT[t][c][k] * 1(|w[k][c][t]| <= 1)
+__global__ void wgrad_finish_kernel(const float* __restrict__ dwT,
+                                    const float* __restrict__ w,
+                                    float* __restrict__ dw, int C, int K) {
+  GRID_STRIDE(i, (int64_t)9 * C * K) {
+    int t = int(i % 9);
+    int64_t rem = i / 9;
+    int c = int(rem % C);
+    int k = int(rem / C);
+    float wv = w[i];
+    float v = dwT[((int64_t)t * C + c) * K + k];
+    dw[i] = (wv <= 1.f && wv >= -1.f) ? v : 0.f;
+  }
+}
+
+extern "C" void bdbnn_wgrad_finish(const float* dwT, const float* w,
+                                   float* dw, int C, int K,
+                                   hipStream_t stream) {
+  int64_t total = (int64_t)9 * C * K;
+  int blocks = (int)bd_min<int64_t>((total + 255) / 256, 8192);
+  wgrad_finish_kernel<<<blocks, 256, 0, stream>>>(dwT, w, dw, C, K);
+}

@@ -129,14 +129,15 @@ class BinaryConvFunction(torch.autograd.Function):
                     and stride == 1 and padding == 1 and kh == 3
                     and nat.dgrad2_supported(H, W, C, K))
             if use2:
+                # dgrad: halo implicit GEMM, clip-STE mask fused
                 wd = nat.dgrad_weight_decode(wp, alpha, C)
                 dx = nat.conv_dgrad2(g, wd, mp, C)
-                xb = nat.decode_packed(xp, C, bf16)
-                wb = nat.weight_decode(wp, alpha, C, bf16)
-                dwb = torch.ops.aten.convolution_backward(
-                    g, xb, wb, None, [stride, stride], [padding, padding],
-                    [1, 1], False, [0, 0], 1, [False, True, False])[1]
-                dw = nat.ste_mask_mul(dwb, w, 0, 0.0, 0.0)
+                # wgrad: all-9-tap block GEMM straight from the sign
+                # BITS (the dense +-1 activation tensor is never
+                # materialized), then transpose + |w|<=1 mask in one pass
+                xcp = nat.repack_cplane(xp, C, W)
+                dwT = nat.conv_wgrad2(g, xcp, C)
+                dw = nat.wgrad_finish(dwT, w.float())
                 return (dx, dw.to(w.dtype), None, None, None, None, None,
                         None)
             xb = nat.decode_packed(xp, C, bf16)

@@ -699,3 +699,95 @@ def test_packed_backward_uses_mfma_dgrad():
     assert torch.allclose(dx2, dx1, atol=1e-2, rtol=1e-2), \
         (dx2 - dx1).abs().max().item()
     assert torch.allclose(dw2, dw1, atol=1e-2, rtol=1e-2)
+
+
+# ---------------- MFMA wgrad v2 (hot path) ----------------
+
+@pytest.mark.parametrize("N,C,H,K", [
+    (2, 64, 56, 64),
+    (2, 128, 28, 128),
+    (2, 256, 14, 256),
+    (4, 512, 7, 512),
+    (1, 64, 9, 64),      # W < Wp (dummy columns)
+    (3, 64, 30, 64),     # H % rc != 0, odd N
+])
+def test_conv_wgrad2_matches_reference(N, C, H, K):
+    torch.manual_seed(41)
+    nat = _nat()
+    x = torch.randn(N, C, H, H, device="cuda")
+    g = _cl(torch.randn(N, K, H, H, device="cuda", dtype=torch.bfloat16))
+    xp = nat.sign_pack_nhwc(_cl(x))
+    xcp = nat.repack_cplane(xp, C, H)
+    dwT = nat.conv_wgrad2(g, xcp, C)           # [9][C][K] fp32
+    xb = binsign(x).to(torch.bfloat16)
+    ref = torch.ops.aten.convolution_backward(
+        g.float(), _cl(xb.float()),
+        torch.empty(K, C, 3, 3, device="cuda"), None,
+        [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+        [False, True, False])[1]               # [K][C][3][3]
+    got = dwT.permute(2, 1, 0).reshape(K, C, 3, 3)
+    err = (got - ref).abs().max().item()
+    # fp32 atomic accumulation of bf16 products vs MIOpen's order: small
+    # relative slack on sums over N*H*W terms
+    assert torch.allclose(got, ref, atol=2.0, rtol=2e-2), err
+
+
+def test_repack_cplane_bits():
+    torch.manual_seed(42)
+    nat = _nat()
+    N, C, H, W = 2, 64, 5, 9
+    x = torch.randn(N, C, H, W, device="cuda")
+    xp = nat.sign_pack_nhwc(_cl(x))
+    xcp = nat.repack_cplane(xp, C, W).cpu().numpy().astype("uint64")
+    xs = x.permute(0, 2, 3, 1).cpu()
+    for c in range(0, C, 17):
+        for n in range(N):
+            for y in range(H):
+                row = int(xcp[c][n * H + y])
+                for xx in range(64):
+                    bit = (row >> xx) & 1
+                    want = (1 if xs[n, y, xx, c].item() >= 0 else 0) \
+                        if xx < W else 0
+                    assert bit == want, (c, n, y, xx)
+
+
+def test_wgrad_finish_transpose_and_mask():
+    torch.manual_seed(43)
+    nat = _nat()
+    K, C = 64, 64
+    dwT = torch.randn(9, C, K, device="cuda")
+    w = (torch.randn(K, C, 3, 3, device="cuda") * 1.2)
+    dw = nat.wgrad_finish(dwT, w)
+    ref = dwT.permute(2, 1, 0).reshape(K, C, 3, 3) * (w.abs() <= 1).float()
+    assert torch.equal(dw, ref)
+
+
+def test_full_packed_backward_mfma_vs_miopen():
+    """Both dx and dw of the default path vs the BDBNN_MFMA_BWD=0 path."""
+    import bdbnn_amd.ops.binary_conv as bc
+    torch.manual_seed(44)
+    for (N, C, H, K) in [(2, 64, 28, 64), (2, 64, 7, 128)]:
+        x = _cl(torch.randn(N, C, H, H, device="cuda",
+                            dtype=torch.bfloat16))
+        w = torch.randn(K, C, 3, 3, device="cuda")
+
+        def run(flag):
+            old = bc._MFMA_BWD
+            bc._MFMA_BWD = flag
+            try:
+                xl = x.clone().requires_grad_(True)
+                wl = w.clone().requires_grad_(True)
+                out, _, _ = BinaryConvFunction.apply(
+                    xl, wl, 1, 1, "ste", None, None, False)
+                (out.float() * torch.randn_like(out.float())).sum().backward()
+                return xl.grad.float(), wl.grad.float()
+            finally:
+                bc._MFMA_BWD = old
+        torch.manual_seed(45)
+        dx2, dw2 = run(True)
+        torch.manual_seed(45)
+        dx1, dw1 = run(False)
+        assert torch.allclose(dx2, dx1, atol=5e-2, rtol=2e-2), \
+            (dx2 - dx1).abs().max().item()
+        assert torch.allclose(dw2, dw1, atol=5e-1, rtol=2e-2), \
+            (dw2 - dw1).abs().max().item()
