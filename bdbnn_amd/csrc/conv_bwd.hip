@@ -50,18 +50,25 @@ struct Dgrad2Params {
 // One template instantiation per padded-width class.
 //   WP: padded row width; RB: image rows per band; GB: bands per block.
 //   GB*(RB)*WP == 256.
+//
+// Each block is quasi-persistent: it loops over a CONTIGUOUS range of
+// m-tiles (and all g-channel chunks of each), double-buffering the
+// halo, the B tile AND the per-tile tables across the whole
+// (tile, k-chunk) stream — the per-tile startup serialization of the
+// naive one-tile-per-block version (measured 8x off the MFMA floor at
+// 1 block/CU occupancy) disappears, and consecutive tiles' overlapping
+// halo rows stay L2/L1-warm on the same CU.
 template <int WP, int RB, int GB>
 __global__ __launch_bounds__(512, 2) void conv_dgrad2_kernel(
     const __bf16* __restrict__ g, const __bf16* __restrict__ wd,
     const uint32_t* __restrict__ mp, __bf16* __restrict__ dx,
-    Dgrad2Params p, int grid_m) {
+    Dgrad2Params p, int grid_m, int nb_m, int tiles_per_block) {
   constexpr int WH = WP + 2;                  // halo row width
   constexpr int NHE = GB * (RB + 2) * WH;     // halo entries (128 B each)
   constexpr int PIECES = NHE * 8;             // 16-B staging pieces
   constexpr int PPT = (PIECES + 511) / 512;   // pieces per thread
 
-  // XCD-aware bijective remap (8 XCDs with private L2s): contiguous
-  // spatial tiles per XCD so neighbouring bands share L2 lines.
+  // XCD-aware bijective remap (8 XCDs with private L2s)
   int nwg = gridDim.x;
   int wg = blockIdx.x;
   {
@@ -69,51 +76,56 @@ __global__ __launch_bounds__(512, 2) void conv_dgrad2_kernel(
     int xcd = wg % 8, idx = wg / 8;
     wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
   }
-  const int m_blk = wg % grid_m;
-  const int c_blk = wg / grid_m;
+  const int rb_m = wg % nb_m;               // which m-tile RANGE
+  const int c_blk = wg / nb_m;
   const int c0 = c_blk * DG2_BN;
+  const int tile0 = rb_m * tiles_per_block;
+  const int tile_end = bd_min(tile0 + tiles_per_block, grid_m);
+  const int my_tiles = tile_end - tile0;
+  if (my_tiles <= 0) return;
   const int tid = threadIdx.x;
 
   __shared__ __align__(16) __bf16 halo[2][NHE * DG2_BK];
   __shared__ __align__(16) __bf16 blds[2][DG2_BN * DG2_BK];
-  __shared__ int he_base[NHE];     // g element offset of halo entry, or -1
-  __shared__ int tab_he[DG2_BM];   // he00 of m-row (tap dy=dx=0)
-  __shared__ int tab_out[DG2_BM];  // output pixel index, or -1
-  __shared__ unsigned char tab_x7[DG2_BM];  // x & 7 (swizzle key base)
+  __shared__ int he_base[2][NHE];   // g pixel index of halo entry, or -1
+  __shared__ int tab_he[2][DG2_BM];   // he00 of m-row (tap dy=dx=0)
+  __shared__ int tab_out[2][DG2_BM];  // output pixel index, or -1
+  __shared__ unsigned char tab_x7[2][DG2_BM];  // x & 7 (swizzle key base)
 
-  // ---- per-block tables ----
-  const int slot0 = m_blk * GB;
-  for (int he = tid; he < NHE; he += 512) {
-    int band = he / ((RB + 2) * WH);
-    int rem = he - band * ((RB + 2) * WH);
-    int hr = rem / WH, hx = rem - hr * WH;
-    int slot = slot0 + band;
-    int base = -1;
-    if (slot < p.total_slots) {
-      int n = slot / p.bands_per_image;
-      int r0 = (slot - n * p.bands_per_image) * RB;
-      int y = r0 + hr - 1, x = hx - 1;
-      if (y >= 0 && y < p.H && x >= 0 && x < p.W)
-        base = (n * p.H + y) * p.W + x;   // pixel index (x K at use)
-    }
-    he_base[he] = base;
+#define BUILD_TABLES(tile, tb)                                            \
+  {                                                                       \
+    const int slot0 = (tile)*GB;                                          \
+    for (int he = tid; he < NHE; he += 512) {                             \
+      int band = he / ((RB + 2) * WH);                                    \
+      int rem = he - band * ((RB + 2) * WH);                              \
+      int hr = rem / WH, hx = rem - hr * WH;                              \
+      int slot = slot0 + band;                                            \
+      int base = -1;                                                      \
+      if (slot < p.total_slots) {                                         \
+        int n = slot / p.bands_per_image;                                 \
+        int r0 = (slot - n * p.bands_per_image) * RB;                     \
+        int y = r0 + hr - 1, x = hx - 1;                                  \
+        if (y >= 0 && y < p.H && x >= 0 && x < p.W)                       \
+          base = (n * p.H + y) * p.W + x;                                 \
+      }                                                                   \
+      he_base[tb][he] = base;                                             \
+    }                                                                     \
+    for (int m = tid; m < DG2_BM; m += 512) {                             \
+      int band = m / (RB * WP);                                           \
+      int rem = m - band * (RB * WP);                                     \
+      int rloc = rem / WP, x = rem - rloc * WP;                           \
+      int slot = slot0 + band;                                            \
+      tab_he[tb][m] = band * (RB + 2) * WH + rloc * WH + x;               \
+      tab_x7[tb][m] = (unsigned char)(x & 7);                             \
+      int out = -1;                                                       \
+      if (slot < p.total_slots) {                                         \
+        int n = slot / p.bands_per_image;                                 \
+        int y = (slot - n * p.bands_per_image) * RB + rloc;               \
+        if (y < p.H && x < p.W) out = (n * p.H + y) * p.W + x;            \
+      }                                                                   \
+      tab_out[tb][m] = out;                                               \
+    }                                                                     \
   }
-  for (int m = tid; m < DG2_BM; m += 512) {
-    int band = m / (RB * WP);
-    int rem = m - band * (RB * WP);
-    int rloc = rem / WP, x = rem - rloc * WP;
-    int slot = slot0 + band;
-    tab_he[m] = band * (RB + 2) * WH + rloc * WH + x;
-    tab_x7[m] = (unsigned char)(x & 7);
-    int out = -1;
-    if (slot < p.total_slots) {
-      int n = slot / p.bands_per_image;
-      int y = (slot - n * p.bands_per_image) * RB + rloc;
-      if (y < p.H && x < p.W) out = (n * p.H + y) * p.W + x;
-    }
-    tab_out[m] = out;
-  }
-  __syncthreads();
 
   // ---- wave decomposition: 8 waves as 4(M) x 2(N) ----
   const int wid = tid >> 6;
@@ -123,22 +135,27 @@ __global__ __launch_bounds__(512, 2) void conv_dgrad2_kernel(
   const int lrow = lane & 31;
   const int lk8 = lane >> 5;             // which 8-element k-half
 
-  // per-lane A addressing for its two 32-row M-fragments
   const int mA0 = wm0 + lrow, mA1 = wm0 + 32 + lrow;
-  const int heA0 = tab_he[mA0], heA1 = tab_he[mA1];
-  const int x7A0 = tab_x7[mA0], x7A1 = tab_x7[mA1];
-  // B addressing: lane's c row
   const int cB = wc + lrow;
 
-  f32x16 acc0 = {0.f}, acc1 = {0.f};
+  // per-lane A addressing (reloaded at each tile switch)
+  int heA0, heA1, x7A0, x7A1;
+#define LOAD_LANE_TABS(tb)                                                \
+  {                                                                       \
+    heA0 = tab_he[tb][mA0];                                               \
+    heA1 = tab_he[tb][mA1];                                               \
+    x7A0 = tab_x7[tb][mA0];                                               \
+    x7A1 = tab_x7[tb][mA1];                                               \
+  }
+
+  f32x16 acc0, acc1;
 #pragma unroll
   for (int i = 0; i < 16; ++i) { acc0[i] = 0.f; acc1[i] = 0.f; }
 
-  // ---- staging (registers -> LDS, write-late) ----
   uint4 hreg[PPT];
   uint4 breg;
 
-#define HALO_LOAD(k0)                                                     \
+#define HALO_LOAD(tb, k0)                                                 \
   {                                                                       \
     _Pragma("unroll")                                                     \
     for (int it = 0; it < PPT; ++it) {                                    \
@@ -146,7 +163,7 @@ __global__ __launch_bounds__(512, 2) void conv_dgrad2_kernel(
       uint4 v{0, 0, 0, 0};                                                \
       if (i < PIECES) {                                                   \
         int he = i >> 3, k8 = i & 7;                                      \
-        int base = he_base[he];                                           \
+        int base = he_base[tb][he];                                       \
         if (base >= 0)                                                    \
           v = *(const uint4*)(g + (int64_t)base * p.K + (k0) + k8 * 8);   \
       }                                                                   \
@@ -168,7 +185,6 @@ __global__ __launch_bounds__(512, 2) void conv_dgrad2_kernel(
     }                                                                     \
   }
 
-  // B tile for tap t: wd[t][c0+c][k0..k0+63]; thread loads one 16-B piece
   const int b_c = tid >> 3;        // 0..63
   const int b_k8 = tid & 7;        // 0..7
 #define B_LOAD(t, k0)                                                     \
@@ -177,28 +193,66 @@ __global__ __launch_bounds__(512, 2) void conv_dgrad2_kernel(
 #define B_WRITE(buf)                                                      \
   *(uint4*)&blds[buf][b_c * DG2_BK + ((b_k8 ^ (b_c & 7)) << 3)] = breg;
 
-  HALO_LOAD(0);
+  // epilogue: clip-STE mask from the packed bitplane, bf16 store.
+  // C/D layout: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5).
+  const int ccol = c0 + wc + lrow;
+  const int cw_word = ccol >> 5;
+  const int cbit = ccol & 31;
+#define EPILOGUE(tb)                                                      \
+  {                                                                       \
+    _Pragma("unroll")                                                     \
+    for (int half = 0; half < 2; ++half) {                                \
+      const f32x16& acc = half ? acc1 : acc0;                             \
+      const int mbase = wm0 + half * 32 + 4 * lk8;                        \
+      _Pragma("unroll")                                                   \
+      for (int reg = 0; reg < 16; ++reg) {                                \
+        int m = mbase + (reg & 3) + 8 * (reg >> 2);                       \
+        int pix = tab_out[tb][m];                                         \
+        if (pix < 0) continue;                                            \
+        uint32_t word = mp[(int64_t)pix * p.CW + cw_word];                \
+        float v = (word >> cbit) & 1 ? acc[reg] : 0.f;                    \
+        uint16_t h = f32_to_bf16(v);                                      \
+        *(uint16_t*)(dx + (int64_t)pix * p.C + ccol) = h;                 \
+      }                                                                   \
+    }                                                                     \
+  }
+
+  // ---- (tile, k-chunk) stream with cross-tile prefetch ----
+  const int n_k0 = p.K / DG2_BK;
+  const int S = my_tiles * n_k0;
+
+  BUILD_TABLES(tile0, 0);
+  __syncthreads();
+  LOAD_LANE_TABS(0);
+  HALO_LOAD(0, 0);
   HALO_WRITE(0);
   B_LOAD(0, 0);
   B_WRITE(0);
   __syncthreads();
 
-  int hb = 0, bb = 0;
-  const int n_k0 = p.K / DG2_BK;
-  for (int ki = 0; ki < n_k0; ++ki) {
+  int hb = 0, bb = 0, tb = 0;
+  for (int s = 0; s < S; ++s) {
+    const int ki = s % n_k0;
     const int k0 = ki * DG2_BK;
-    const bool more = ki + 1 < n_k0;
-    if (more) HALO_LOAD(k0 + DG2_BK);   // issue early, write at tap 8
+    const bool more = s + 1 < S;
+    const bool tile_switch = (ki == n_k0 - 1);  // next s starts a new tile
+    const int nki = tile_switch ? 0 : ki + 1;
 #pragma unroll
     for (int t = 0; t < 9; ++t) {
       const int dy = t / 3, dxt = t - dy * 3;
-      // prefetch next B tile (next tap, or tap 0 of the next chunk)
       if (t < 8) {
         B_LOAD(t + 1, k0);
       } else if (more) {
-        B_LOAD(0, k0 + DG2_BK);
+        B_LOAD(0, nki * DG2_BK);
       }
-      // this tap's A base addresses (element offsets into halo[hb])
+      if (t == 0 && more && tile_switch) {
+        // build the NEXT tile's tables into the other buffer; the tap-0
+        // barrier below publishes them before HALO_LOAD reads them
+        BUILD_TABLES(tile0 + (s + 1) / n_k0, tb ^ 1);
+      }
+      if (t == 1 && more) {
+        HALO_LOAD(tile_switch ? tb ^ 1 : tb, nki * DG2_BK);
+      }
       const int heT0 = (heA0 + dy * WH + dxt) * DG2_BK;
       const int heT1 = (heA1 + dy * WH + dxt) * DG2_BK;
       const int keyA0 = ((x7A0 + dxt) & 7) ^ lk8;
@@ -206,7 +260,6 @@ __global__ __launch_bounds__(512, 2) void conv_dgrad2_kernel(
       const int keyB = (cB & 7) ^ lk8;
 #pragma unroll
       for (int kk = 0; kk < 4; ++kk) {
-        // kk8 index of this 16-deep step's k-half = kk*2 ^ swizzle key
         bf16x8 a0 = *(const bf16x8*)&halo[hb][heT0 + (((kk << 1) ^ keyA0) << 3)];
         bf16x8 a1 = *(const bf16x8*)&halo[hb][heT1 + (((kk << 1) ^ keyA1) << 3)];
         bf16x8 b = *(const bf16x8*)&blds[bb][cB * DG2_BK + (((kk << 1) ^ keyB) << 3)];
@@ -221,29 +274,21 @@ __global__ __launch_bounds__(512, 2) void conv_dgrad2_kernel(
       }
     }
     hb ^= 1;
-  }
-
-  // ---- epilogue: clip-STE mask from the packed bitplane, bf16 store ----
-  // C/D layout (guide section 3): col = lane&31, row = (reg&3) + 8*(reg>>2)
-  // + 4*(lane>>5).
-  const int ccol = c0 + wc + lrow;
-  const int cw_word = ccol >> 5;              // uniform per 32-lane half
-  const int cbit = ccol & 31;
+    if (tile_switch) {
+      EPILOGUE(tb);
 #pragma unroll
-  for (int half = 0; half < 2; ++half) {
-    const f32x16& acc = half ? acc1 : acc0;
-    const int mbase = wm0 + half * 32 + 4 * lk8;
-#pragma unroll
-    for (int reg = 0; reg < 16; ++reg) {
-      int m = mbase + (reg & 3) + 8 * (reg >> 2);
-      int pix = tab_out[m];
-      if (pix < 0) continue;
-      uint32_t word = mp[(int64_t)pix * p.CW + cw_word];
-      float v = (word >> cbit) & 1 ? acc[reg] : 0.f;
-      uint16_t h = f32_to_bf16(v);
-      *(uint16_t*)(dx + (int64_t)pix * p.C + ccol) = h;
+      for (int i = 0; i < 16; ++i) { acc0[i] = 0.f; acc1[i] = 0.f; }
+      tb ^= 1;
+      LOAD_LANE_TABS(tb);
     }
   }
+#undef BUILD_TABLES
+#undef LOAD_LANE_TABS
+#undef HALO_LOAD
+#undef HALO_WRITE
+#undef B_LOAD
+#undef B_WRITE
+#undef EPILOGUE
 }
 
 // ---------------- host launcher ----------------
@@ -261,9 +306,16 @@ extern "C" int bdbnn_conv_dgrad2(const void* g, const void* wd,
     p.bands_per_image = (H + (RBV)-1) / (RBV);                            \
     p.total_slots = N * p.bands_per_image;                                \
     grid_m = (p.total_slots + (GBV)-1) / (GBV);                           \
-    dim3 grid(grid_m * (C / DG2_BN));                                     \
+    int n_ctile = C / DG2_BN;                                             \
+    /* 1 block/CU resident (LDS-bound): ~2 ranges per CU for tail      */ \
+    /* balance, contiguous m-tiles per block for halo L2 reuse         */ \
+    int nb_m = bd_min(grid_m, bd_min(512, 768 / n_ctile));                \
+    int tpb = (grid_m + nb_m - 1) / nb_m;                                 \
+    nb_m = (grid_m + tpb - 1) / tpb;                                      \
+    dim3 grid(nb_m * n_ctile);                                            \
     conv_dgrad2_kernel<WPV, RBV, GBV><<<grid, 512, 0, stream>>>(          \
-        (const __bf16*)g, (const __bf16*)wd, mp, (__bf16*)dx, p, grid_m); \
+        (const __bf16*)g, (const __bf16*)wd, mp, (__bf16*)dx, p, grid_m,  \
+        nb_m, tpb);                                                       \
     return 0;                                                             \
   }
   if (W <= 8) {
@@ -494,7 +546,7 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
     __syncthreads();                  // X + gT[gb] ready for all waves
 
 #pragma unroll
-    for (int msl = 0; msl < 2; ++msl) {
+    for (int msl = 0; msl < 4; ++msl) {   // 4 x 16-m steps = its 64-m half
       const int m16 = ms_grp * 64 + msl * 16;
       // B-frag: 8 m at fixed k from gT
       bf16x8 bfrag = *(const bf16x8*)&gT[gb][b_k * GSTRIDE + m16 +

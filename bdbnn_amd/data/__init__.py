@@ -1,5 +1,6 @@
 from .loaders import (
     SyntheticImageDataset,
+    LearnableSyntheticDataset,
     CIFAR10Dataset,
     CIFAR100Dataset,
     ImageFolderDataset,
