@@ -73,7 +73,7 @@ void bdbnn_prelu_bwd(const void*, const void*, const float*, void*, float*,
 void bdbnn_bn_stats(const void*, float*, float*, int64_t, int, bool,
                     hipStream_t);
 void bdbnn_bn_finalize(const float*, const float*, float*, float*, float*,
-                       float*, int, float, float, float, hipStream_t);
+                       float*, int, float, float, float, int, hipStream_t);
 void bdbnn_bn_act_fwd(const void*, const void*, const float*, const float*,
                       const float*, const float*, const float*, void*, void*,
                       int64_t, int, int, bool, hipStream_t);
@@ -339,8 +339,8 @@ std::vector<at::Tensor> xnor_conv_fwd(
   float *s1p = nullptr, *s2p = nullptr;
   if (want_stats) {
     auto fopt = xp.options().dtype(at::kFloat);
-    s1 = at::empty({K}, fopt);
-    s2 = at::empty({K}, fopt);
+    s1 = at::empty({32, K}, fopt);   // 32-way sliced partial sums
+    s2 = at::empty({32, K}, fopt);
     s1p = s1.data_ptr<float>();
     s2p = s2.data_ptr<float>();
   }
@@ -491,12 +491,16 @@ std::vector<at::Tensor> bn_act_fwd_train(
   bool bf16 = is_bf16(xc);
   auto fopt = xc.options().dtype(at::kFloat);
   at::Tensor s1, s2;
+  int nslice = 1;
   auto mean = at::empty({C}, fopt);
   auto invstd = at::empty({C}, fopt);
   if (pre_s1.has_value()) {
     // stats already accumulated by the producing conv's epilogue
+    // (possibly as [nslice][C] partial sums — finalize folds them)
     s1 = *pre_s1;
     s2 = *pre_s2;
+    nslice = (int)(s1.numel() / C);
+    TORCH_CHECK(nslice * C == s1.numel(), "bn: stats shape mismatch");
   } else {
     s1 = at::empty({C}, fopt);
     s2 = at::empty({C}, fopt);
@@ -509,7 +513,7 @@ std::vector<at::Tensor> bn_act_fwd_train(
                   ? running_var->data_ptr<float>() : nullptr;
   bdbnn_bn_finalize(s1.data_ptr<float>(), s2.data_ptr<float>(),
                     mean.data_ptr<float>(), invstd.data_ptr<float>(), rm, rv,
-                    C, (float)(n / C), (float)momentum, (float)eps,
+                    C, (float)(n / C), (float)momentum, (float)eps, nslice,
                     cur_stream());
   auto out = at::empty_like(xc, xc.options(), at::MemoryFormat::ChannelsLast);
   // PReLU backward needs the true pre-activation z; ReLU only needs its

@@ -50,17 +50,25 @@ __global__ void bn_stats_kernel(const T* __restrict__ x,
 }
 
 // ---- finalize: mean/invstd + running-stat update ----
+// nslice > 1: s1/s2 are [nslice][C] partial sums (the conv epilogue's
+// 32-way sliced accumulators) — summed here, where it costs nothing.
 __global__ void bn_finalize_kernel(const float* __restrict__ s1,
                                    const float* __restrict__ s2,
                                    float* __restrict__ mean,
                                    float* __restrict__ invstd,
                                    float* __restrict__ running_mean,
                                    float* __restrict__ running_var,
-                                   int C, float n, float momentum, float eps) {
+                                   int C, float n, float momentum, float eps,
+                                   int nslice) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float m = s1[c] / n;
-  float var = fmaxf(s2[c] / n - m * m, 0.f);
+  float a1 = 0.f, a2 = 0.f;
+  for (int j = 0; j < nslice; ++j) {
+    a1 += s1[j * C + c];
+    a2 += s2[j * C + c];
+  }
+  float m = a1 / n;
+  float var = fmaxf(a2 / n - m * m, 0.f);
   mean[c] = m;
   invstd[c] = rsqrtf(var + eps);
   if (running_mean != nullptr) {
@@ -342,9 +350,10 @@ extern "C" void bdbnn_bn_finalize(const float* s1, const float* s2,
                                   float* mean, float* invstd,
                                   float* running_mean, float* running_var,
                                   int C, float n, float momentum, float eps,
-                                  hipStream_t stream) {
+                                  int nslice, hipStream_t stream) {
   bn_finalize_kernel<<<(C + 255) / 256, 256, 0, stream>>>(
-      s1, s2, mean, invstd, running_mean, running_var, C, n, momentum, eps);
+      s1, s2, mean, invstd, running_mean, running_var, C, n, momentum, eps,
+      nslice);
 }
 
 extern "C" void bdbnn_bn_act_fwd(const void* x, const void* skip,

@@ -386,6 +386,10 @@ extern "C" void bdbnn_dgrad_wdec(const uint32_t* wp, const float* alpha,
 // into dwT[9][C][K]; wgrad_finish transposes to [K][C][3][3] and
 // applies the |w|<=1 STE mask in the same pass.
 //
+// A chunk is GB image-BANDS of RB rows each (GB*RB*WP == 128), so small
+// images (7x7 at WP=8) fill a whole band instead of 49/128 of a flat
+// chunk — each band has its own 2-row halo in X.
+//
 // Constraints: 3x3/s1/p1, C % 64 == 0, K % 64 == 0, W <= 64, g bf16
 // channels_last.
 
@@ -395,19 +399,19 @@ extern "C" void bdbnn_dgrad_wdec(const uint32_t* wp, const float* alpha,
 
 struct Wgrad2Params {
   int N, H, W, C, K;
-  int chunks_per_image;     // ceil(H / rc)
-  int total_chunks;         // N * chunks_per_image
+  int bands_per_image;      // ceil(H / RB)
+  int total_slots;          // N * bands_per_image
+  int total_chunks;         // ceil(total_slots / GB)
   int split;                // chunk stride (== blocks per (c,k) tile)
   int nh_rows;              // N * H  (xcp row count)
 };
 
-template <int WP>
+template <int WP, int RB, int GB>
 __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
     const __bf16* __restrict__ g, const uint64_t* __restrict__ xcp,
     float* __restrict__ dwT, Wgrad2Params p, int grid_ck) {
-  constexpr int RC = WG2_CHUNK / WP;      // image rows per chunk
-  constexpr int XROWS = RC + 2;           // halo rows decoded per chunk
-  constexpr int XSTRIDE = XROWS * WP + 8; // +16 B pad: bank-conflict-free
+  constexpr int XROWS = RB + 2;           // halo rows per band
+  constexpr int XSTRIDE = GB * XROWS * WP + 8;  // +16 B pad: conflict-free
   constexpr int GSTRIDE = WG2_CHUNK + 8;  // gT row stride (elements)
   constexpr int LGWP = WP == 8 ? 3 : WP == 16 ? 4 : WP == 32 ? 5 : 6;
 
@@ -419,11 +423,17 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
   const int k_blk = ck / (p.C / WG2_BC);
   const int c0 = c_blk * WG2_BC, k0 = k_blk * WG2_BK;
 
-  // X: [3 copies][64 c] rows of XROWS*WP (+pad) elements
+  // X: [3 copies][64 c] rows of GB*XROWS*WP (+pad) elements
   __shared__ __align__(16) __bf16 X[3 * WG2_BC * XSTRIDE];
   __shared__ __align__(16) __bf16 gT[2][WG2_BK * GSTRIDE];
   __shared__ __align__(16) __bf16 lut[256][8];
   __shared__ float red[4][32][32];        // m-split combine scratch
+  // raw xcp row bits, double-buffered: the global u64 loads prefetch
+  // into registers during the PREVIOUS chunk's MFMA phase, so X_DECODE
+  // reads them at LDS latency instead of L2/HBM latency
+  constexpr int NBITS = WG2_BC * GB * XROWS;
+  __shared__ uint64_t xbits[2][NBITS];
+  __shared__ unsigned char xvalid[2][NBITS];  // 0 = y-pad row (decode to 0)
 
   // ---- byte -> 8 x (+-1) LUT (bit 1 <=> x >= 0 <=> +1) ----
   for (int b = tid; b < 256; b += 512) {
@@ -448,8 +458,8 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
   //       and scatter-writes them transposed.
   const int sg_m = tid >> 2;              // 0..127
   const int sg_k16 = (tid & 3) * 16;      // k offset of its 16 elements
-  // X decode: targets (copy dxs, c, row, 8-x block), grid-strided.
-  constexpr int XTGT = 3 * WG2_BC * XROWS * (WP / 8);
+  // X decode: targets (copy dxs, c, band, row, 8-x block), grid-strided.
+  constexpr int XTGT = 3 * WG2_BC * GB * XROWS * (WP / 8);
 
   const int n_chunks = (p.total_chunks - split_id + p.split - 1) / p.split;
   int ch = split_id;
@@ -458,17 +468,21 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
 
 #define G_LOAD(chunk)                                                     \
   {                                                                       \
-    int n = (chunk) / p.chunks_per_image;                                 \
-    int y0 = ((chunk) - n * p.chunks_per_image) * RC;                     \
-    int y = y0 + (sg_m >> LGWP);                                          \
-    int x = sg_m & (WP - 1);                                              \
+    int band = sg_m / (RB * WP);                                          \
+    int rem = sg_m - band * (RB * WP);                                    \
+    int slot = (chunk)*GB + band;                                         \
+    int x = rem & (WP - 1);                                               \
     uint4 z{0, 0, 0, 0};                                                  \
     greg[0] = z; greg[1] = z;                                             \
-    if (x < p.W && y < p.H) {                                             \
-      const __bf16* src =                                                 \
-          g + ((int64_t)(n * p.H + y) * p.W + x) * p.K + k0 + sg_k16;     \
-      greg[0] = *(const uint4*)src;                                       \
-      greg[1] = *(const uint4*)(src + 8);                                 \
+    if (slot < p.total_slots && x < p.W) {                                \
+      int n = slot / p.bands_per_image;                                   \
+      int y = (slot - n * p.bands_per_image) * RB + (rem >> LGWP);        \
+      if (y < p.H) {                                                      \
+        const __bf16* src =                                               \
+            g + ((int64_t)(n * p.H + y) * p.W + x) * p.K + k0 + sg_k16;   \
+        greg[0] = *(const uint4*)src;                                     \
+        greg[1] = *(const uint4*)(src + 8);                               \
+      }                                                                   \
     }                                                                     \
   }
 
@@ -481,11 +495,54 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
     }                                                                     \
   }
 
-  // ---- X decode for chunk ch (LDS-resident after barrier) ----
-#define X_DECODE(chunk)                                                   \
+  // ---- raw-bits prefetch: global u64 loads issue during the previous
+  // chunk's MFMA phase and land in LDS at the chunk boundary ----
+  constexpr int BPT = (NBITS + 511) / 512;   // bits rows per thread
+  uint64_t bits_reg[BPT];
+  unsigned char valid_reg[BPT];
+#define BITS_LOAD(chunk)                                                  \
   {                                                                       \
-    int n = (chunk) / p.chunks_per_image;                                 \
-    int y0 = ((chunk) - n * p.chunks_per_image) * RC;                     \
+    _Pragma("unroll")                                                     \
+    for (int it = 0; it < BPT; ++it) {                                    \
+      int i = tid + it * 512;                                             \
+      uint64_t b = 0;                                                     \
+      unsigned char v = 0;                                                \
+      if (i < NBITS) {                                                    \
+        int row = i % XROWS;                                              \
+        int rem = i / XROWS;                                              \
+        int band = rem % GB;                                              \
+        int c = rem / GB;                                                 \
+        int slot = (chunk)*GB + band;                                     \
+        if (slot < p.total_slots) {                                       \
+          int n = slot / p.bands_per_image;                               \
+          int y = (slot - n * p.bands_per_image) * RB + row - 1;          \
+          if (y >= 0 && y < p.H) {                                        \
+            b = xcp[(int64_t)(c0 + c) * p.nh_rows + n * p.H + y];         \
+            v = 1;                                                        \
+          }                                                               \
+        }                                                                 \
+      }                                                                   \
+      bits_reg[it] = b;                                                   \
+      valid_reg[it] = v;                                                  \
+    }                                                                     \
+  }
+#define BITS_WRITE(buf)                                                   \
+  {                                                                       \
+    _Pragma("unroll")                                                     \
+    for (int it = 0; it < BPT; ++it) {                                    \
+      int i = tid + it * 512;                                             \
+      if (i < NBITS) {                                                    \
+        xbits[buf][i] = bits_reg[it];                                     \
+        xvalid[buf][i] = valid_reg[it];                                   \
+      }                                                                   \
+    }                                                                     \
+  }
+
+  // ---- X decode for one chunk, from the LDS-resident bits; the x-pad
+  // edge zeros (X0[x=0], X2[x=W-1]) are folded into the target write ----
+  const int eb8 = (p.W - 1) >> 3, eel = (p.W - 1) & 7;
+#define X_DECODE(buf)                                                     \
+  {                                                                       \
     _Pragma("unroll 2")                                                   \
     for (int it = 0; it < (XTGT + 511) / 512; ++it) {                     \
       int i = tid + it * 512;                                             \
@@ -494,40 +551,31 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
         int rem = i / (WP / 8);                                           \
         int row = rem % XROWS;                                            \
         rem /= XROWS;                                                     \
+        int band = rem % GB;                                              \
+        rem /= GB;                                                        \
         int c = rem % WG2_BC;                                             \
         int dxs = rem / WG2_BC;                                           \
-        int y = y0 + row - 1;                                             \
+        int bi = (c * GB + band) * XROWS + row;                           \
+        uint64_t bits = xbits[buf][bi];                                   \
+        /* kw shift: X_dxs[x] = xb[x + dxs - 1] */                        \
+        uint64_t sh = dxs == 0 ? (bits << 1) : (bits >> (dxs - 1));       \
+        unsigned byte = (unsigned)(sh >> (8 * xb8)) & 0xffu;              \
         uint4 vv{0, 0, 0, 0};                                             \
-        if (y >= 0 && y < p.H) {                                          \
-          /* y-pad rows stay ZERO (a 0-bit byte would decode to -1) */    \
-          uint64_t bits =                                                 \
-              xcp[(int64_t)(c0 + c) * p.nh_rows + n * p.H + y];           \
-          /* kw shift: X_dxs[x] = xb[x + dxs - 1] */                      \
-          uint64_t sh = dxs == 0 ? (bits << 1) : (bits >> (dxs - 1));     \
-          unsigned byte = (unsigned)(sh >> (8 * xb8)) & 0xffu;            \
+        if (xvalid[buf][bi]) {                                            \
           vv = *(const uint4*)&lut[byte][0];                              \
+          if (dxs == 0 && xb8 == 0) ((uint16_t*)&vv)[0] = 0;              \
+          if (dxs == 2 && xb8 == eb8) ((uint16_t*)&vv)[eel] = 0;          \
         }                                                                 \
-        *(uint4*)&X[(dxs * WG2_BC + c) * XSTRIDE + row * WP + xb8 * 8] =  \
-            vv;                                                           \
+        *(uint4*)&X[(dxs * WG2_BC + c) * XSTRIDE +                        \
+                    (band * XROWS + row) * WP + xb8 * 8] = vv;            \
       }                                                                   \
-    }                                                                     \
-    __syncthreads();                                                      \
-    /* edge zeros: X0[x=0] (left pad) and X2[x=W-1] (right pad) */        \
-    for (int i = tid; i < 2 * WG2_BC * XROWS; i += 512) {                 \
-      int row = i % XROWS;                                                \
-      int rem = i / XROWS;                                                \
-      int c = rem % WG2_BC;                                               \
-      int which = rem / WG2_BC;                                           \
-      if (which == 0)                                                     \
-        X[(0 * WG2_BC + c) * XSTRIDE + row * WP + 0] = (__bf16)0.f;       \
-      else                                                                \
-        X[(2 * WG2_BC + c) * XSTRIDE + row * WP + (p.W - 1)] =            \
-            (__bf16)0.f;                                                  \
     }                                                                     \
   }
 
   if (n_chunks > 0) {
+    BITS_LOAD(ch);
     G_LOAD(ch);
+    BITS_WRITE(0);
     G_WRITE(0);
   }
 
@@ -540,10 +588,13 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
   int gb = 0;
   for (int ci = 0; ci < n_chunks; ++ci) {
     const bool more = ci + 1 < n_chunks;
-    // decode this chunk's X (also serves as the gT visibility barrier)
-    X_DECODE(ch);
-    if (more) G_LOAD(ch + p.split);   // issue next g early
-    __syncthreads();                  // X + gT[gb] ready for all waves
+    __syncthreads();                  // xbits[gb] + gT[gb] visible; X free
+    X_DECODE(gb);                     // LDS-latency decode into X
+    if (more) {
+      BITS_LOAD(ch + p.split);        // issue next chunk's loads early:
+      G_LOAD(ch + p.split);           // they land during the MFMA phase
+    }
+    __syncthreads();                  // X ready for all waves
 
 #pragma unroll
     for (int msl = 0; msl < 4; ++msl) {   // 4 x 16-m steps = its 64-m half
@@ -551,22 +602,27 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
       // B-frag: 8 m at fixed k from gT
       bf16x8 bfrag = *(const bf16x8*)&gT[gb][b_k * GSTRIDE + m16 +
                                            lhalf * 8];
-      // A-frags per tap from X: row (rl + dy), x-run
+      // A-frags per tap from X: band-local row (+ dy), x-run
       const int mstart = m16 + lhalf * 8;
       const int rl = mstart >> LGWP;
+      const int band = rl / RB, rloc = rl - band * RB;
       const int xs = mstart & (WP - 1);
 #pragma unroll
       for (int t = 0; t < 9; ++t) {
         const int dy = t / 3, dxs = t - dy * 3;
         bf16x8 afrag = *(const bf16x8*)&X[(dxs * WG2_BC + a_c) * XSTRIDE +
-                                          (rl + dy) * WP + xs];
+                                          (band * XROWS + rloc + dy) * WP +
+                                          xs];
         acc[t] =
             __builtin_amdgcn_mfma_f32_32x32x16_bf16(afrag, bfrag, acc[t],
                                                     0, 0, 0);
       }
     }
     __syncthreads();                  // all reads of X/gT[gb] done
-    if (more) G_WRITE(gb ^ 1);
+    if (more) {
+      BITS_WRITE(gb ^ 1);
+      G_WRITE(gb ^ 1);
+    }
     gb ^= 1;
     ch += p.split;
   }
@@ -607,23 +663,26 @@ extern "C" int bdbnn_conv_wgrad2(const void* g, const uint64_t* xcp,
   p.N = N; p.H = H; p.W = W; p.C = C; p.K = K;
   p.nh_rows = N * H;
   int grid_ck = (C / WG2_BC) * (K / WG2_BK);
-  int split = (384 + grid_ck - 1) / grid_ck;
+  // LDS (~140-160 KB/block) admits ONE 512-thread block per CU: round
+  // the m-split so the grid is a whole multiple of 256 CUs (384 blocks
+  // = 1.5 dispatch rounds would idle half the chip for half the time)
+  int split = (256 + grid_ck - 1) / grid_ck;
   p.split = split;
-#define WLAUNCH(WPV)                                                      \
+#define WLAUNCH(WPV, RBV, GBV)                                            \
   {                                                                       \
-    constexpr int rc = WG2_CHUNK / WPV;                                   \
-    p.chunks_per_image = (H + rc - 1) / rc;                               \
-    p.total_chunks = N * p.chunks_per_image;                              \
+    p.bands_per_image = (H + (RBV)-1) / (RBV);                            \
+    p.total_slots = N * p.bands_per_image;                                \
+    p.total_chunks = (p.total_slots + (GBV)-1) / (GBV);                   \
     if (p.split > p.total_chunks) p.split = p.total_chunks;               \
     dim3 grid(grid_ck * p.split);                                         \
-    conv_wgrad2_kernel<WPV><<<grid, 512, 0, stream>>>(                    \
+    conv_wgrad2_kernel<WPV, RBV, GBV><<<grid, 512, 0, stream>>>(          \
         (const __bf16*)g, xcp, dwT, p, grid_ck);                          \
     return 0;                                                             \
   }
-  if (W <= 8) WLAUNCH(8)
-  else if (W <= 16) WLAUNCH(16)
-  else if (W <= 32) WLAUNCH(32)
-  else WLAUNCH(64)
+  if (W <= 8) WLAUNCH(8, 8, 2)
+  else if (W <= 16) WLAUNCH(16, 8, 1)
+  else if (W <= 32) WLAUNCH(32, 4, 1)
+  else WLAUNCH(64, 2, 1)
 #undef WLAUNCH
 }
 

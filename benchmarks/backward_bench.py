@@ -65,8 +65,15 @@ def main():
         d_mi = timeit(lambda: torch.ops.aten.convolution_backward(
             g, xb, wb, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
             [True, False, False])[0])
+        def wgrad2():
+            # everything the v2 wgrad path runs per backward: bit repack,
+            # all-9-tap block GEMM, transpose + STE-mask finish
+            xcp = nat.repack_cplane(xp, C, H)
+            dwT = nat.conv_wgrad2(g, xcp, C)
+            return nat.wgrad_finish(dwT, w)
+
         try:
-            w2 = timeit(lambda: nat.conv_wgrad2(g, xp, C))
+            w2 = timeit(wgrad2)
         except (AttributeError, RuntimeError):
             w2 = float("nan")
         w_mi = timeit(lambda: torch.ops.aten.convolution_backward(

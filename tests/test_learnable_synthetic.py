@@ -54,10 +54,14 @@ def test_short_training_learns_and_kurtosis_converges():
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     out = os.path.join(repo, "gpurun_out", "acc_test_log.jsonl")
     os.makedirs(os.path.dirname(out), exist_ok=True)
+    # recipe calibrated on the CPU oracle path: identical settings reach
+    # ~77% val top-1 there (binary nets need the higher lr to move in a
+    # 10-epoch budget; lr 0.05 + noise 0.5 stays near chance on BOTH
+    # paths, so a tight budget tests recipe, not kernels)
     r = subprocess.run(
-        [sys.executable, "benchmarks/accuracy_run.py", "--epochs", "4",
-         "--train-size", "8192", "--val-size", "1024", "--noise", "0.5",
-         "--out", out],
+        [sys.executable, "benchmarks/accuracy_run.py", "--epochs", "10",
+         "--train-size", "8192", "--val-size", "1024", "--noise", "0.3",
+         "--lr", "0.3", "--out", out],
         cwd=repo, capture_output=True, text=True, timeout=1200)
     assert r.returncode == 0, r.stdout + r.stderr
     summary = [json.loads(l)["summary"] for l in open(out)
