@@ -60,9 +60,10 @@ void bdbnn_dgrad_wdec(const uint32_t*, const float*, void*, int, int,
                       hipStream_t);
 int bdbnn_conv_wgrad2(const void*, const uint64_t*, float*, int, int, int,
                       int, int, hipStream_t);
+int bdbnn_wgrad2_nslab(int, int, int, int, int);
 void bdbnn_repack_cplane(const uint32_t*, uint64_t*, int, int, int, int,
                          hipStream_t);
-void bdbnn_wgrad_finish(const float*, const float*, float*, int, int,
+void bdbnn_wgrad_finish(const float*, const float*, float*, int, int, int,
                         hipStream_t);
 void bdbnn_conv_wgrad(const void*, const uint32_t*, float*, int, int, int,
                       int, int, int, hipStream_t);
@@ -76,7 +77,8 @@ void bdbnn_bn_finalize(const float*, const float*, float*, float*, float*,
                        float*, int, float, float, float, int, hipStream_t);
 void bdbnn_bn_act_fwd(const void*, const void*, const float*, const float*,
                       const float*, const float*, const float*, void*, void*,
-                      int64_t, int, int, bool, hipStream_t);
+                      int64_t, int, int, uint32_t*, uint32_t*, bool,
+                      hipStream_t);
 void bdbnn_bn_act_bwd_reduce(const void*, const void*, const void*,
                              const float*, const float*, const float*,
                              float*, int64_t, int, int, bool, hipStream_t);
@@ -483,7 +485,7 @@ std::vector<at::Tensor> bn_act_fwd_train(
     c10::optional<at::Tensor> running_mean,
     c10::optional<at::Tensor> running_var, double momentum, double eps,
     int64_t act_kind, const c10::optional<at::Tensor>& pre_s1,
-    const c10::optional<at::Tensor>& pre_s2) {
+    const c10::optional<at::Tensor>& pre_s2, bool want_pack) {
   auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
   int C = (int)x.size(1);
   TORCH_CHECK(C <= 1024, "fused bn: C <= 1024");
@@ -539,10 +541,24 @@ std::vector<at::Tensor> bn_act_fwd_train(
   const float* a_ptr = nullptr;
   if (a.has_value()) { af = a->contiguous().to(at::kFloat);
                        a_ptr = af.data_ptr<float>(); }
+  // consumer-conv sign/mask pack fused into the epilogue (the next
+  // binary conv then skips its own pack read pass; csrc/bn_act.hip)
+  at::Tensor xpk, mpk;
+  uint32_t *xpk_p = nullptr, *mpk_p = nullptr;
+  if (want_pack) {
+    TORCH_CHECK(C % 32 == 0, "bn pack fusion needs C % 32 == 0");
+    int CW = C / 32;
+    auto iopt = xc.options().dtype(at::kInt);
+    xpk = at::empty({x.size(0), x.size(2), x.size(3), CW}, iopt);
+    mpk = at::empty({x.size(0), x.size(2), x.size(3), CW}, iopt);
+    xpk_p = (uint32_t*)xpk.data_ptr<int>();
+    mpk_p = (uint32_t*)mpk.data_ptr<int>();
+  }
   bdbnn_bn_act_fwd(xc.data_ptr(), skip_ptr, mean.data_ptr<float>(),
                    invstd.data_ptr<float>(), gf.data_ptr<float>(),
                    bf.data_ptr<float>(), a_ptr, out.data_ptr(), zptr,
-                   n, C, (int)act_kind, bf16, cur_stream());
+                   n, C, (int)act_kind, xpk_p, mpk_p, bf16, cur_stream());
+  if (want_pack) return {out, z, mean, invstd, xpk, mpk};
   return {out, z, mean, invstd};
 }
 
@@ -699,7 +715,10 @@ at::Tensor conv_wgrad2(const at::Tensor& g, const at::Tensor& xcp,
               "conv_wgrad2: c-plane bitplanes [C][N*H]");
   TORCH_CHECK(C % 64 == 0 && K % 64 == 0 && W <= 64,
               "conv_wgrad2: unsupported shape");
-  auto dwT = at::zeros({9, C, K}, g.options().dtype(at::kFloat));
+  int nslab = bdbnn_wgrad2_nslab(N, H, W, (int)C, K);
+  TORCH_CHECK(nslab > 0, "conv_wgrad2: unsupported shape");
+  // per-m-split-block partial slabs; every word is written (no memset)
+  auto dwT = at::empty({nslab, 9, C, K}, g.options().dtype(at::kFloat));
   int rc = bdbnn_conv_wgrad2(g.data_ptr(),
                              (const uint64_t*)xcp.data_ptr<int64_t>(),
                              dwT.data_ptr<float>(), N, H, W, (int)C, K,
@@ -713,14 +732,14 @@ at::Tensor wgrad_finish(const at::Tensor& dwT, const at::Tensor& w) {
                   w.size(3) == 3 && w.scalar_type() == at::kFloat,
               "wgrad_finish: fp32 [K][C][3][3] latent weights");
   int K = (int)w.size(0), C = (int)w.size(1);
-  TORCH_CHECK(dwT.dim() == 3 && dwT.size(0) == 9 && dwT.size(1) == C &&
-                  dwT.size(2) == K && dwT.scalar_type() == at::kFloat,
-              "wgrad_finish: dwT [9][C][K] fp32");
+  TORCH_CHECK(dwT.dim() == 4 && dwT.size(1) == 9 && dwT.size(2) == C &&
+                  dwT.size(3) == K && dwT.scalar_type() == at::kFloat,
+              "wgrad_finish: dwT [nslab][9][C][K] fp32");
   auto wc = w.contiguous();
   auto dw = at::empty_like(wc);
   bdbnn_wgrad_finish(dwT.contiguous().data_ptr<float>(),
                      wc.data_ptr<float>(), dw.data_ptr<float>(), C, K,
-                     cur_stream());
+                     (int)dwT.size(0), cur_stream());
   return dw;
 }
 

@@ -78,8 +78,15 @@ __global__ void bn_finalize_kernel(const float* __restrict__ s1,
   }
 }
 
-// ---- pass 2: normalize + add + act ----
-template <typename T, int ACT>
+// ---- pass 2: normalize + add + act (+ next conv's sign/mask pack) ----
+// PACK: the consuming binary conv's sign + clip-STE-mask bitplanes are
+// assembled in the epilogue from the ROUNDED stored values (bit-exact
+// with csrc/pack.hip sign_mask_pack_kernel on the written tensor), so
+// that conv never re-reads the activation to pack it.  Each thread owns
+// 8 consecutive channels -> one byte of each plane; the 4 lanes of a
+// 32-channel word are consecutive (C % 32 == 0 => threads-per-row % 4
+// == 0) and combine with intra-wave shuffles, lane (tid&3)==0 stores.
+template <typename T, int ACT, bool PACK>
 __global__ void bn_act_fwd_kernel(const T* __restrict__ x,
                                   const T* __restrict__ skip,
                                   const float* __restrict__ mean,
@@ -88,7 +95,9 @@ __global__ void bn_act_fwd_kernel(const T* __restrict__ x,
                                   const float* __restrict__ beta,
                                   const float* __restrict__ a,
                                   T* __restrict__ out, T* __restrict__ zout,
-                                  int64_t n_pix, int C) {
+                                  int64_t n_pix, int C,
+                                  uint32_t* __restrict__ xpk,
+                                  uint32_t* __restrict__ mpk) {
   ChanMap m = chan_map8(C);
   float sc[8], sh[8], av[8];
 #pragma unroll
@@ -98,6 +107,9 @@ __global__ void bn_act_fwd_kernel(const T* __restrict__ x,
     sh[j] = beta[cc] - mean[cc] * sc[j];
     av[j] = (ACT == 1) ? a[cc] : 0.f;
   }
+  const int sub = threadIdx.x & 3;         // byte slot within the word
+  const int64_t cw = m.c0 >> 5;            // word column (lane sub == 0)
+  const int CW = C >> 5;
   float v[8], s[8], z[8], o[8];
   for (int64_t p = m.p0; p < n_pix; p += m.pstep) {
     int64_t i = p * C + m.c0;
@@ -113,6 +125,27 @@ __global__ void bn_act_fwd_kernel(const T* __restrict__ x,
     }
     store8(out, i, o);
     if (zout != nullptr) store8(zout, i, z);
+    if constexpr (PACK) {
+      unsigned both = 0;   // sign byte | mask byte << 8
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float r = (sizeof(T) == 2) ? bf16_to_f32(f32_to_bf16(o[j])) : o[j];
+        both |= (r >= 0.f ? 1u : 0u) << j;
+        both |= (fabsf(r) <= 1.f ? 1u : 0u) << (8 + j);
+      }
+      unsigned b1 = __shfl_down(both, 1);
+      unsigned b2 = __shfl_down(both, 2);
+      unsigned b3 = __shfl_down(both, 3);
+      if (sub == 0) {
+        uint32_t sw = (both & 0xffu) | ((b1 & 0xffu) << 8) |
+                      ((b2 & 0xffu) << 16) | ((b3 & 0xffu) << 24);
+        uint32_t mw = ((both >> 8) & 0xffu) | (((b1 >> 8) & 0xffu) << 8) |
+                      (((b2 >> 8) & 0xffu) << 16) |
+                      (((b3 >> 8) & 0xffu) << 24);
+        xpk[p * CW + cw] = sw;
+        mpk[p * CW + cw] = mw;
+      }
+    }
   }
 }
 
@@ -260,18 +293,23 @@ static void launch_fwd(const void* x, const void* skip, const float* mean,
                        const float* invstd, const float* gamma,
                        const float* beta, const float* a, void* out,
                        void* zout, int64_t n_pix, int C, int act_kind,
-                       int grid, hipStream_t stream) {
+                       uint32_t* xpk, uint32_t* mpk, int grid,
+                       hipStream_t stream) {
   auto X = (const T*)x; auto S = (const T*)skip;
   auto O = (T*)out; auto Z = (T*)zout;
-  if (act_kind == 1)
-    bn_act_fwd_kernel<T, 1><<<grid, 256, 0, stream>>>(
-        X, S, mean, invstd, gamma, beta, a, O, Z, n_pix, C);
-  else if (act_kind == 2)
-    bn_act_fwd_kernel<T, 2><<<grid, 256, 0, stream>>>(
-        X, S, mean, invstd, gamma, beta, a, O, Z, n_pix, C);
-  else
-    bn_act_fwd_kernel<T, 0><<<grid, 256, 0, stream>>>(
-        X, S, mean, invstd, gamma, beta, a, O, Z, n_pix, C);
+#define BN_FWD(ACT)                                                       \
+  {                                                                       \
+    if (xpk != nullptr)                                                   \
+      bn_act_fwd_kernel<T, ACT, true><<<grid, 256, 0, stream>>>(          \
+          X, S, mean, invstd, gamma, beta, a, O, Z, n_pix, C, xpk, mpk);  \
+    else                                                                  \
+      bn_act_fwd_kernel<T, ACT, false><<<grid, 256, 0, stream>>>(         \
+          X, S, mean, invstd, gamma, beta, a, O, Z, n_pix, C, xpk, mpk);  \
+  }
+  if (act_kind == 1) BN_FWD(1)
+  else if (act_kind == 2) BN_FWD(2)
+  else BN_FWD(0)
+#undef BN_FWD
 }
 
 template <typename T>
@@ -360,16 +398,17 @@ extern "C" void bdbnn_bn_act_fwd(const void* x, const void* skip,
                                  const float* mean, const float* invstd,
                                  const float* gamma, const float* beta,
                                  const float* a, void* out, void* zout,
-                                 int64_t n, int C, int act_kind, bool bf16,
+                                 int64_t n, int C, int act_kind,
+                                 uint32_t* xpk, uint32_t* mpk, bool bf16,
                                  hipStream_t stream) {
   int64_t n_pix = n / C;
   int grid = grid_pix8(n_pix, C);
   if (bf16)
     launch_fwd<uint16_t>(x, skip, mean, invstd, gamma, beta, a, out, zout,
-                         n_pix, C, act_kind, grid, stream);
+                         n_pix, C, act_kind, xpk, mpk, grid, stream);
   else
     launch_fwd<float>(x, skip, mean, invstd, gamma, beta, a, out, zout,
-                      n_pix, C, act_kind, grid, stream);
+                      n_pix, C, act_kind, xpk, mpk, grid, stream);
 }
 
 extern "C" void bdbnn_bn_act_bwd_reduce(const void* dy, const void* z,

@@ -406,14 +406,24 @@ struct Wgrad2Params {
   int nh_rows;              // N * H  (xcp row count)
 };
 
+// Block geometry: each (c,k) tile is m-split over `split` blocks; a
+// block owns a CONTIGUOUS chunk range, so (for GB==1) consecutive
+// chunks are consecutive row-bands of one image and the X decode only
+// produces the RB NEW rows per chunk — the 2 halo rows ride a row RING
+// (plane = (y+1) mod XROWS) decoded by the previous chunk.  Partials go
+// to a per-block slab dwT[split][9][C][K] (plain coalesced stores);
+// wgrad_finish sums the slabs — no end-of-kernel atomic storm on the
+// small dwT (C=K=64: 256 blocks x 37k words of atomicAdd measured as a
+// multi-10us tail).
 template <int WP, int RB, int GB>
 __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
     const __bf16* __restrict__ g, const uint64_t* __restrict__ xcp,
     float* __restrict__ dwT, Wgrad2Params p, int grid_ck) {
-  constexpr int XROWS = RB + 2;           // halo rows per band
+  constexpr int XROWS = RB + 2;           // ring planes (halo rows) per band
   constexpr int XSTRIDE = GB * XROWS * WP + 8;  // +16 B pad: conflict-free
   constexpr int GSTRIDE = WG2_CHUNK + 8;  // gT row stride (elements)
   constexpr int LGWP = WP == 8 ? 3 : WP == 16 ? 4 : WP == 32 ? 5 : 6;
+  constexpr bool RING = (GB == 1);        // GB>1: bands are whole images
 
   const int tid = threadIdx.x;
   int wg = blockIdx.x;
@@ -425,15 +435,15 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
 
   // X: [3 copies][64 c] rows of GB*XROWS*WP (+pad) elements
   __shared__ __align__(16) __bf16 X[3 * WG2_BC * XSTRIDE];
-  __shared__ __align__(16) __bf16 gT[2][WG2_BK * GSTRIDE];
+  __shared__ __align__(16) __bf16 gT[WG2_BK * GSTRIDE];
   __shared__ __align__(16) __bf16 lut[256][8];
   __shared__ float red[4][32][32];        // m-split combine scratch
-  // raw xcp row bits, double-buffered: the global u64 loads prefetch
-  // into registers during the PREVIOUS chunk's MFMA phase, so X_DECODE
-  // reads them at LDS latency instead of L2/HBM latency
+  // raw xcp row bits: the global u64 loads prefetch into registers
+  // during the previous chunk's MFMA phase and land here at the chunk
+  // boundary, so X_DECODE reads them at LDS latency
   constexpr int NBITS = WG2_BC * GB * XROWS;
-  __shared__ uint64_t xbits[2][NBITS];
-  __shared__ unsigned char xvalid[2][NBITS];  // 0 = y-pad row (decode to 0)
+  __shared__ uint64_t xbits[NBITS];
+  __shared__ unsigned char xvalid[NBITS];  // 0 = pad row (decode to 0)
 
   // ---- byte -> 8 x (+-1) LUT (bit 1 <=> x >= 0 <=> +1) ----
   for (int b = tid; b < 256; b += 512) {
@@ -453,69 +463,84 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
 #pragma unroll
     for (int i = 0; i < 16; ++i) acc[t][i] = 0.f;
 
+  // contiguous chunk range of this split block
+  const int cpb = (p.total_chunks + p.split - 1) / p.split;
+  const int ch_lo = split_id * cpb;
+  const int ch_hi = bd_min(ch_lo + cpb, p.total_chunks);
+  const int n_chunks = ch_hi - ch_lo;   // may be 0: slab still written
+  int ch = ch_lo;
+
   // staging assignments (fixed per thread)
-  //   gT: thread loads 16 g elements (two 16-B pieces along k at fixed m)
-  //       and scatter-writes them transposed.
-  const int sg_m = tid >> 2;              // 0..127
-  const int sg_k16 = (tid & 3) * 16;      // k offset of its 16 elements
-  // X decode: targets (copy dxs, c, band, row, 8-x block), grid-strided.
-  constexpr int XTGT = 3 * WG2_BC * GB * XROWS * (WP / 8);
-
-  const int n_chunks = (p.total_chunks - split_id + p.split - 1) / p.split;
-  int ch = split_id;
-
+  //   gT: thread loads 8 k-elements of TWO adjacent m rows and writes 8
+  //   paired b32 stores (half the LDS-write instructions of a b16
+  //   scatter transpose).
+  const int sg_m2 = (tid & 63) * 2;       // m pair base
+  const int sg_k8 = (tid >> 6) * 8;       // its 8 k elements
   uint4 greg[2];
 
-#define G_LOAD(chunk)                                                     \
+#define G_ADDR(chunk, m, out_ptr)                                         \
   {                                                                       \
-    int band = sg_m / (RB * WP);                                          \
-    int rem = sg_m - band * (RB * WP);                                    \
+    int band = (m) / (RB * WP);                                           \
+    int rem = (m)-band * (RB * WP);                                       \
     int slot = (chunk)*GB + band;                                         \
     int x = rem & (WP - 1);                                               \
-    uint4 z{0, 0, 0, 0};                                                  \
-    greg[0] = z; greg[1] = z;                                             \
+    out_ptr = nullptr;                                                    \
     if (slot < p.total_slots && x < p.W) {                                \
       int n = slot / p.bands_per_image;                                   \
       int y = (slot - n * p.bands_per_image) * RB + (rem >> LGWP);        \
-      if (y < p.H) {                                                      \
-        const __bf16* src =                                               \
-            g + ((int64_t)(n * p.H + y) * p.W + x) * p.K + k0 + sg_k16;   \
-        greg[0] = *(const uint4*)src;                                     \
-        greg[1] = *(const uint4*)(src + 8);                               \
-      }                                                                   \
+      if (y < p.H)                                                        \
+        out_ptr = g + ((int64_t)(n * p.H + y) * p.W + x) * p.K + k0 +     \
+                  sg_k8;                                                  \
     }                                                                     \
   }
 
-#define G_WRITE(buf)                                                      \
+#define G_LOAD(chunk)                                                     \
+  {                                                                       \
+    const __bf16* src0;                                                   \
+    const __bf16* src1;                                                   \
+    G_ADDR(chunk, sg_m2, src0);                                           \
+    G_ADDR(chunk, sg_m2 + 1, src1);                                       \
+    uint4 z{0, 0, 0, 0};                                                  \
+    greg[0] = src0 ? *(const uint4*)src0 : z;                             \
+    greg[1] = src1 ? *(const uint4*)src1 : z;                             \
+  }
+
+#define G_WRITE()                                                         \
   {                                                                       \
     _Pragma("unroll")                                                     \
-    for (int j = 0; j < 16; ++j) {                                        \
-      __bf16 v = ((const __bf16*)greg)[j];                                \
-      gT[buf][(sg_k16 + j) * GSTRIDE + sg_m] = v;                         \
+    for (int j = 0; j < 8; ++j) {                                         \
+      uint32_t pair = (uint32_t)((const uint16_t*)greg)[j] |              \
+                      ((uint32_t)((const uint16_t*)&greg[1])[j] << 16);   \
+      *(uint32_t*)&gT[(sg_k8 + j) * GSTRIDE + sg_m2] = pair;              \
     }                                                                     \
   }
 
-  // ---- raw-bits prefetch: global u64 loads issue during the previous
-  // chunk's MFMA phase and land in LDS at the chunk boundary ----
-  constexpr int BPT = (NBITS + 511) / 512;   // bits rows per thread
+  // ---- raw-bits prefetch ----
+  // rows are addressed by WINDOW position w in [r0, r0+NR): global row
+  // y = y0 + w - 1, ring plane = RING ? (y0+w) % XROWS : w.  A fresh
+  // chunk decodes the whole window (r0=0, NR=XROWS); a continuation
+  // chunk (same image, next band) only the RB new bottom rows (r0=2).
+  constexpr int BPT = (NBITS + 511) / 512;
   uint64_t bits_reg[BPT];
   unsigned char valid_reg[BPT];
-#define BITS_LOAD(chunk)                                                  \
+  int nbits_cur = 0;
+#define BITS_LOAD(chunk, r0, NR)                                          \
   {                                                                       \
+    nbits_cur = WG2_BC * GB * (NR);                                       \
     _Pragma("unroll")                                                     \
     for (int it = 0; it < BPT; ++it) {                                    \
       int i = tid + it * 512;                                             \
       uint64_t b = 0;                                                     \
       unsigned char v = 0;                                                \
-      if (i < NBITS) {                                                    \
-        int row = i % XROWS;                                              \
-        int rem = i / XROWS;                                              \
+      if (i < nbits_cur) {                                                \
+        int w = (r0) + i % (NR);                                          \
+        int rem = i / (NR);                                               \
         int band = rem % GB;                                              \
         int c = rem / GB;                                                 \
         int slot = (chunk)*GB + band;                                     \
         if (slot < p.total_slots) {                                       \
           int n = slot / p.bands_per_image;                               \
-          int y = (slot - n * p.bands_per_image) * RB + row - 1;          \
+          int y = (slot - n * p.bands_per_image) * RB + w - 1;            \
           if (y >= 0 && y < p.H) {                                        \
             b = xcp[(int64_t)(c0 + c) * p.nh_rows + n * p.H + y];         \
             v = 1;                                                        \
@@ -526,57 +551,83 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
       valid_reg[it] = v;                                                  \
     }                                                                     \
   }
-#define BITS_WRITE(buf)                                                   \
+#define BITS_WRITE(chunk, r0, NR)                                         \
   {                                                                       \
     _Pragma("unroll")                                                     \
     for (int it = 0; it < BPT; ++it) {                                    \
       int i = tid + it * 512;                                             \
-      if (i < NBITS) {                                                    \
-        xbits[buf][i] = bits_reg[it];                                     \
-        xvalid[buf][i] = valid_reg[it];                                   \
+      if (i < nbits_cur) {                                                \
+        int w = (r0) + i % (NR);                                          \
+        int rem = i / (NR);                                               \
+        int band = rem % GB;                                              \
+        int c = rem / GB;                                                 \
+        int plane = w;                                                    \
+        if (RING) {                                                       \
+          int slot = (chunk)*GB + band;                                   \
+          int y0 = (slot - (slot / p.bands_per_image) *                   \
+                              p.bands_per_image) * RB;                    \
+          plane = (y0 + w) % XROWS;                                       \
+        }                                                                 \
+        int bi = (c * GB + band) * XROWS + plane;                         \
+        xbits[bi] = bits_reg[it];                                         \
+        xvalid[bi] = valid_reg[it];                                       \
       }                                                                   \
     }                                                                     \
   }
 
-  // ---- X decode for one chunk, from the LDS-resident bits; the x-pad
+  // ---- X decode for one chunk from the LDS-resident bits; the x-pad
   // edge zeros (X0[x=0], X2[x=W-1]) are folded into the target write ----
   const int eb8 = (p.W - 1) >> 3, eel = (p.W - 1) & 7;
-#define X_DECODE(buf)                                                     \
+#define X_DECODE(chunk, r0, NR)                                           \
   {                                                                       \
+    const int xtgt = 3 * WG2_BC * GB * (NR) * (WP / 8);                   \
     _Pragma("unroll 2")                                                   \
-    for (int it = 0; it < (XTGT + 511) / 512; ++it) {                     \
+    for (int it = 0; it < (3 * WG2_BC * GB * XROWS * (WP / 8) + 511) /    \
+                              512; ++it) {                                \
       int i = tid + it * 512;                                             \
-      if (i < XTGT) {                                                     \
+      if (i < xtgt) {                                                     \
         int xb8 = i % (WP / 8);                                           \
         int rem = i / (WP / 8);                                           \
-        int row = rem % XROWS;                                            \
-        rem /= XROWS;                                                     \
+        int w = (r0) + rem % (NR);                                        \
+        rem /= (NR);                                                      \
         int band = rem % GB;                                              \
         rem /= GB;                                                        \
         int c = rem % WG2_BC;                                             \
         int dxs = rem / WG2_BC;                                           \
-        int bi = (c * GB + band) * XROWS + row;                           \
-        uint64_t bits = xbits[buf][bi];                                   \
+        int plane = w;                                                    \
+        if (RING) {                                                       \
+          int slot = (chunk)*GB + band;                                   \
+          int y0 = (slot - (slot / p.bands_per_image) *                   \
+                              p.bands_per_image) * RB;                    \
+          plane = (y0 + w) % XROWS;                                       \
+        }                                                                 \
+        int bi = (c * GB + band) * XROWS + plane;                         \
+        uint64_t bits = xbits[bi];                                        \
         /* kw shift: X_dxs[x] = xb[x + dxs - 1] */                        \
         uint64_t sh = dxs == 0 ? (bits << 1) : (bits >> (dxs - 1));       \
         unsigned byte = (unsigned)(sh >> (8 * xb8)) & 0xffu;              \
         uint4 vv{0, 0, 0, 0};                                             \
-        if (xvalid[buf][bi]) {                                            \
+        if (xvalid[bi]) {                                                 \
           vv = *(const uint4*)&lut[byte][0];                              \
           if (dxs == 0 && xb8 == 0) ((uint16_t*)&vv)[0] = 0;              \
           if (dxs == 2 && xb8 == eb8) ((uint16_t*)&vv)[eel] = 0;          \
         }                                                                 \
         *(uint4*)&X[(dxs * WG2_BC + c) * XSTRIDE +                        \
-                    (band * XROWS + row) * WP + xb8 * 8] = vv;            \
+                    (band * XROWS + plane) * WP + xb8 * 8] = vv;          \
       }                                                                   \
     }                                                                     \
   }
 
+  // continuation = same image's next band (GB==1 ranges walk bands in
+  // order; a new image or the range start re-decodes the full window)
+#define IS_CONT(chunk) \
+  (RING && (chunk) > ch_lo && ((chunk) % p.bands_per_image) != 0)
+
   if (n_chunks > 0) {
-    BITS_LOAD(ch);
+    BITS_LOAD(ch, 0, XROWS);
     G_LOAD(ch);
-    BITS_WRITE(0);
-    G_WRITE(0);
+    BITS_WRITE(ch, 0, XROWS);
+    G_WRITE();
   }
 
   // per-lane fragment bases
@@ -585,14 +636,23 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
   // B (gT): lane row k = qk*32 + lrow
   const int b_k = qk * 32 + lrow;
 
-  int gb = 0;
   for (int ci = 0; ci < n_chunks; ++ci) {
     const bool more = ci + 1 < n_chunks;
-    __syncthreads();                  // xbits[gb] + gT[gb] visible; X free
-    X_DECODE(gb);                     // LDS-latency decode into X
+    const bool cont = IS_CONT(ch);
+    const bool cont_n = IS_CONT(ch + 1);
+    // y0 of this chunk modulo the ring (0 when !RING: y0 == 0 there)
+    int y0m = 0;
+    if (RING) {
+      int y0 = (ch - (ch / p.bands_per_image) * p.bands_per_image) * RB;
+      y0m = y0 % XROWS;
+    }
+    __syncthreads();                  // xbits + gT visible; X free
+    if (cont) X_DECODE(ch, 2, RB)
+    else X_DECODE(ch, 0, XROWS)
     if (more) {
-      BITS_LOAD(ch + p.split);        // issue next chunk's loads early:
-      G_LOAD(ch + p.split);           // they land during the MFMA phase
+      if (cont_n) BITS_LOAD(ch + 1, 2, RB)
+      else BITS_LOAD(ch + 1, 0, XROWS)
+      G_LOAD(ch + 1);                 // lands during the MFMA phase
     }
     __syncthreads();                  // X ready for all waves
 
@@ -600,36 +660,43 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
     for (int msl = 0; msl < 4; ++msl) {   // 4 x 16-m steps = its 64-m half
       const int m16 = ms_grp * 64 + msl * 16;
       // B-frag: 8 m at fixed k from gT
-      bf16x8 bfrag = *(const bf16x8*)&gT[gb][b_k * GSTRIDE + m16 +
-                                           lhalf * 8];
-      // A-frags per tap from X: band-local row (+ dy), x-run
+      bf16x8 bfrag = *(const bf16x8*)&gT[b_k * GSTRIDE + m16 + lhalf * 8];
+      // A-frags per tap from X: ring plane (rloc + dy), x-run
       const int mstart = m16 + lhalf * 8;
       const int rl = mstart >> LGWP;
       const int band = rl / RB, rloc = rl - band * RB;
       const int xs = mstart & (WP - 1);
+      int pl[3];
+#pragma unroll
+      for (int dy = 0; dy < 3; ++dy) {
+        int v = RING ? y0m + rloc + dy : rloc + dy;
+        if (RING && v >= XROWS) v -= XROWS;
+        if (RING && v >= XROWS) v -= XROWS;
+        pl[dy] = band * XROWS + v;
+      }
 #pragma unroll
       for (int t = 0; t < 9; ++t) {
         const int dy = t / 3, dxs = t - dy * 3;
         bf16x8 afrag = *(const bf16x8*)&X[(dxs * WG2_BC + a_c) * XSTRIDE +
-                                          (band * XROWS + rloc + dy) * WP +
-                                          xs];
+                                          pl[dy] * WP + xs];
         acc[t] =
             __builtin_amdgcn_mfma_f32_32x32x16_bf16(afrag, bfrag, acc[t],
                                                     0, 0, 0);
       }
     }
-    __syncthreads();                  // all reads of X/gT[gb] done
+    __syncthreads();                  // all reads of X/gT done
     if (more) {
-      BITS_WRITE(gb ^ 1);
-      G_WRITE(gb ^ 1);
+      if (cont_n) BITS_WRITE(ch + 1, 2, RB)
+      else BITS_WRITE(ch + 1, 0, XROWS)
+      G_WRITE();
     }
-    gb ^= 1;
-    ch += p.split;
+    ++ch;
   }
 
-  // ---- m-split combine + global atomic add ----
+  // ---- m-split combine + plain coalesced slab store ----
   // C/D layout: col j (k) = lane&31, row (c) = (reg&3)+8*(reg>>2)+4*lhalf
   const int quad = wid & 3;
+  float* slab = dwT + (int64_t)split_id * 9 * p.C * p.K;
 #pragma unroll
   for (int t = 0; t < 9; ++t) {
     __syncthreads();
@@ -646,13 +713,29 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
       for (int reg = 0; reg < 16; ++reg) {
         int row = (reg & 3) + 8 * (reg >> 2) + 4 * lhalf;
         float v = acc[t][reg] + red[quad][row][lrow];
-        if (v != 0.f)
-          atomicAdd(&dwT[((int64_t)t * p.C + c0 + qc * 32 + row) * p.K +
-                         k0 + qk * 32 + lrow],
-                    v);
+        slab[((int64_t)t * p.C + c0 + qc * 32 + row) * p.K + k0 +
+             qk * 32 + lrow] = v;
       }
     }
   }
+}
+
+// split (= slab count) mirrored for the host-side dwT allocation
+extern "C" int bdbnn_wgrad2_nslab(int N, int H, int W, int C, int K) {
+  if (C % 64 || K % 64 || W > 64) return -1;
+  int grid_ck = (C / WG2_BC) * (K / WG2_BK);
+  int split = (256 + grid_ck - 1) / grid_ck;
+  int bands, total_chunks;
+  if (W <= 8 && H <= 8) {
+    bands = 1; total_chunks = (N + 1) / 2;
+  } else if (W <= 16) {
+    bands = (H + 7) / 8; total_chunks = N * bands;
+  } else if (W <= 32) {
+    bands = (H + 3) / 4; total_chunks = N * bands;
+  } else {
+    bands = (H + 1) / 2; total_chunks = N * bands;
+  }
+  return split < total_chunks ? split : total_chunks;
 }
 
 extern "C" int bdbnn_conv_wgrad2(const void* g, const uint64_t* xcp,
@@ -663,7 +746,7 @@ extern "C" int bdbnn_conv_wgrad2(const void* g, const uint64_t* xcp,
   p.N = N; p.H = H; p.W = W; p.C = C; p.K = K;
   p.nh_rows = N * H;
   int grid_ck = (C / WG2_BC) * (K / WG2_BK);
-  // LDS (~140-160 KB/block) admits ONE 512-thread block per CU: round
+  // LDS (~110-145 KB/block) admits ONE 512-thread block per CU: round
   // the m-split so the grid is a whole multiple of 256 CUs (384 blocks
   // = 1.5 dispatch rounds would idle half the chip for half the time)
   int split = (256 + grid_ck - 1) / grid_ck;
@@ -679,7 +762,7 @@ extern "C" int bdbnn_conv_wgrad2(const void* g, const uint64_t* xcp,
         (const __bf16*)g, xcp, dwT, p, grid_ck);                          \
     return 0;                                                             \
   }
-  if (W <= 8) WLAUNCH(8, 8, 2)
+  if (W <= 8 && H <= 8) WLAUNCH(8, 8, 2)
   else if (W <= 16) WLAUNCH(16, 8, 1)
   else if (W <= 32) WLAUNCH(32, 4, 1)
   else WLAUNCH(64, 2, 1)
@@ -722,26 +805,32 @@ extern "C" void bdbnn_repack_cplane(const uint32_t* xp, uint64_t* xcp,
   repack_cplane_kernel<<<blocks, 256, 0, stream>>>(xp, xcp, NH, W, C, CW);
 }
 
-// ---------------- wgrad finish: transpose + STE mask ----------------
-// dw[k][c][t] = dwT[t][c][k] * 1(|w[k][c][t]| <= 1)
+// ---------------- wgrad finish: slab sum + transpose + STE mask ----
+// dw[k][c][t] = sum_s dwT[s][t][c][k] * 1(|w[k][c][t]| <= 1)
+// (the slabs are conv_wgrad2's per-m-split-block partials; summing them
+// here replaces an end-of-kernel atomic storm on the tiny dwT)
 __global__ void wgrad_finish_kernel(const float* __restrict__ dwT,
                                     const float* __restrict__ w,
-                                    float* __restrict__ dw, int C, int K) {
-  GRID_STRIDE(i, (int64_t)9 * C * K) {
+                                    float* __restrict__ dw, int C, int K,
+                                    int nslab) {
+  int64_t stride = (int64_t)9 * C * K;
+  GRID_STRIDE(i, stride) {
     int t = int(i % 9);
     int64_t rem = i / 9;
     int c = int(rem % C);
     int k = int(rem / C);
     float wv = w[i];
-    float v = dwT[((int64_t)t * C + c) * K + k];
+    int64_t j = ((int64_t)t * C + c) * K + k;
+    float v = 0.f;
+    for (int s = 0; s < nslab; ++s) v += dwT[s * stride + j];
     dw[i] = (wv <= 1.f && wv >= -1.f) ? v : 0.f;
   }
 }
 
 extern "C" void bdbnn_wgrad_finish(const float* dwT, const float* w,
-                                   float* dw, int C, int K,
+                                   float* dw, int C, int K, int nslab,
                                    hipStream_t stream) {
   int64_t total = (int64_t)9 * C * K;
   int blocks = (int)bd_min<int64_t>((total + 255) / 256, 8192);
-  wgrad_finish_kernel<<<blocks, 256, 0, stream>>>(dwT, w, dw, C, K);
+  wgrad_finish_kernel<<<blocks, 256, 0, stream>>>(dwT, w, dw, C, K, nslab);
 }

@@ -30,9 +30,9 @@ from ..ops.pool import FusedMaxPool2d
 class FusedDownsample(nn.Sequential):
     """conv + BN downsample with the fused BN kernel (names '0'/'1' kept)."""
 
-    def forward(self, x):
+    def forward(self, x, prepack=None):
         if hasattr(self[0], "forward_with_stats"):
-            out, stats = self[0].forward_with_stats(x)
+            out, stats = self[0].forward_with_stats(x, prepack=prepack)
             return fused_bn_act(out, self[1], stats=stats)
         return fused_bn_act(self[0](x), self[1])
 
@@ -104,12 +104,28 @@ class BiBasicBlock(nn.Module):
         self.act2 = _make_act(act, planes)
 
     def forward(self, x):
-        identity = self.downsample(x) if self.downsample is not None else x
-        o1, st1 = self.conv1.forward_with_stats(x)
-        out = fused_bn_act(o1, self.bn1, self.act1, skip=identity, stats=st1)
-        o2, st2 = self.conv2.forward_with_stats(out)
-        out = fused_bn_act(o2, self.bn2, self.act2, skip=out, stats=st2)
-        return out
+        # accept (x, pack): the previous block's bn2 epilogue packed this
+        # block's input bitplanes (conv1 AND the downsample conv share x)
+        pk_in = None
+        if isinstance(x, tuple):
+            x, pk_in = x
+        if self.downsample is not None:
+            identity = self.downsample(x, prepack=pk_in)
+        else:
+            identity = x
+        o1, st1 = self.conv1.forward_with_stats(x, prepack=pk_in)
+        # only a plain ChannelPReLU tail writes the exact conv2 input in
+        # its epilogue (RPReLU's shifts run after, so no pack there)
+        fuse1 = isinstance(self.act1, ChannelPReLU)
+        r1 = fused_bn_act(o1, self.bn1, self.act1, skip=identity, stats=st1,
+                          pack=fuse1)
+        out, pk1 = r1 if fuse1 else (r1, None)
+        o2, st2 = self.conv2.forward_with_stats(out, prepack=pk1)
+        fuse2 = isinstance(self.act2, ChannelPReLU)
+        r2 = fused_bn_act(o2, self.bn2, self.act2, skip=out, stats=st2,
+                          pack=fuse2)
+        out, pk2 = r2 if fuse2 else (r2, None)
+        return (out, pk2) if pk2 is not None else out
 
 
 class ResNet(nn.Module):
@@ -166,6 +182,8 @@ class ResNet(nn.Module):
         x = self.layer2(x)
         x = self.layer3(x)
         x = self.layer4(x) if hasattr(self, "layer4") else x
+        if isinstance(x, tuple):   # drop the last block's pack hand-off
+            x = x[0]
         x = self.avgpool(x)
         x = torch.flatten(x, 1)
         return self.fc(x)
@@ -253,6 +271,8 @@ class CifarResNet(nn.Module):
         x = self.layer1(x)
         x = self.layer2(x)
         x = self.layer3(x)
+        if isinstance(x, tuple):   # drop the last block's pack hand-off
+            x = x[0]
         x = self.avgpool(x)
         x = torch.flatten(x, 1)
         return self.fc(x)

@@ -18,10 +18,20 @@ class _BinConvBlock(nn.Module):
         self.bn = nn.BatchNorm2d(cout)
         self.act = ChannelPReLU(cout)
         self.pool = nn.MaxPool2d(2) if pool else nn.Identity()
+        self.has_pool = pool
 
     def forward(self, x):
-        out, stats = self.conv.forward_with_stats(x)
-        return self.pool(fused_bn_act(out, self.bn, self.act, stats=stats))
+        # accept (x, pack) from the previous block's fused BN epilogue
+        pk_in = None
+        if isinstance(x, tuple):
+            x, pk_in = x
+        out, stats = self.conv.forward_with_stats(x, prepack=pk_in)
+        if self.has_pool:
+            # pooled output != BN output: a pack would be stale
+            return self.pool(fused_bn_act(out, self.bn, self.act,
+                                          stats=stats))
+        y, pk = fused_bn_act(out, self.bn, self.act, stats=stats, pack=True)
+        return (y, pk) if pk is not None else y
 
 
 class VGGSmall(nn.Module):

@@ -63,7 +63,7 @@ class BinaryConvFunction(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w, stride, padding, act_mode, t, k,
-                want_stats=False):
+                want_stats=False, xp_pre=None, mp_pre=None):
         ctx.stride = stride
         ctx.padding = padding
         ctx.act_mode = act_mode
@@ -72,25 +72,32 @@ class BinaryConvFunction(torch.autograd.Function):
         ctx.packed = False
         if x.is_cuda:
             nat = _C.native_required()
-            xc = x.contiguous(memory_format=torch.channels_last)
             wp, alpha, stab = nat.weight_pack(w)  # bits, alpha[K], S[K][T]
             # (grad mode is always off inside Function.forward; use
             # needs_input_grad to detect inference/no-grad calls)
             if not (ctx.needs_input_grad[0] or ctx.needs_input_grad[1]):
                 # inference/validation: no mask plane, nothing saved
-                xp = nat.sign_pack_nhwc(xc)
+                xp = (xp_pre if xp_pre is not None else nat.sign_pack_nhwc(
+                    x.contiguous(memory_format=torch.channels_last)))
                 ctx.packed = None  # backward must never run
             elif act_mode == "ste" and t is None:
-                # packed fast path: sign + clip-STE mask bitplanes in one
-                # pass; the fp activations are NOT saved — backward works
-                # entirely from the 1-bit planes (32x less read traffic)
-                xp, mp = nat.sign_mask_pack_nhwc(xc)
+                # packed fast path: sign + clip-STE mask bitplanes (either
+                # pre-packed by the producing BN's epilogue, or in one
+                # pass here); the fp activations are NOT saved — backward
+                # works entirely from the 1-bit planes (32x less read
+                # traffic)
+                if xp_pre is not None and mp_pre is not None:
+                    xp, mp = xp_pre, mp_pre
+                else:
+                    xp, mp = nat.sign_mask_pack_nhwc(
+                        x.contiguous(memory_format=torch.channels_last))
                 ctx.packed = True
                 ctx.x_dtype = x.dtype
                 ctx.in_channels = x.shape[1]
                 ctx.save_for_backward(w, xp, mp, wp, alpha)
             else:
-                xp = nat.sign_pack_nhwc(xc)
+                xp = (xp_pre if xp_pre is not None else nat.sign_pack_nhwc(
+                    x.contiguous(memory_format=torch.channels_last)))
                 ctx.save_for_backward(x, w)
             res = nat.xnor_conv_fwd(
                 xp, wp, alpha, stab, x.shape[1], stride, padding,
@@ -139,7 +146,7 @@ class BinaryConvFunction(torch.autograd.Function):
                 dwT = nat.conv_wgrad2(g, xcp, C)
                 dw = nat.wgrad_finish(dwT, w.float())
                 return (dx, dw.to(w.dtype), None, None, None, None, None,
-                        None)
+                        None, None, None)
             xb = nat.decode_packed(xp, C, bf16)
             wb = nat.weight_decode(wp, alpha, C, bf16)
             dxb, dwb = torch.ops.aten.convolution_backward(
@@ -149,7 +156,8 @@ class BinaryConvFunction(torch.autograd.Function):
             dx = nat.mask_mul_packed(dxb, mp, C,
                                      ctx.x_dtype == torch.bfloat16)
             dw = nat.ste_mask_mul(dwb, w, 0, 0.0, 0.0)
-            return dx, dw.to(w.dtype), None, None, None, None, None, None
+            return (dx, dw.to(w.dtype), None, None, None, None, None,
+                    None, None, None)
         x, w = ctx.saved_tensors
         if x.is_cuda:
             nat = _C.native_required()
@@ -177,7 +185,8 @@ class BinaryConvFunction(torch.autograd.Function):
         else:
             dx = dxb * _act_grad_mask(x, ctx.act_mode, ctx.t, ctx.k)
             dw = dwb * (w.abs() <= 1).to(w.dtype)
-        return dx, dw.to(w.dtype), None, None, None, None, None, None
+        return (dx, dw.to(w.dtype), None, None, None, None, None, None,
+                None, None)
 
 
 class _HardBinaryConvBase(nn.Module):
@@ -215,15 +224,28 @@ class _HardBinaryConvBase(nn.Module):
             self.weight.data = self.weight.data.contiguous()
         return self
 
-    def forward(self, x):
+    def _check_prepack(self, x, prepack):
+        """Validate a (xp, mp) pair packed by the producing BN's epilogue
+        against this conv's input; returns (xp, mp) or (None, None)."""
+        if prepack is None or not x.is_cuda:
+            return None, None
+        xp, mp = prepack
+        if (xp.shape[0] == x.shape[0] and xp.shape[1] == x.shape[2]
+                and xp.shape[2] == x.shape[3]
+                and xp.shape[3] * 32 == self.in_channels):
+            return xp, mp
+        return None, None
+
+    def forward(self, x, prepack=None):
         t = float(self.t) if self.t is not None else None
         k = float(self.k) if self.k is not None else None
+        xp, mp = self._check_prepack(x, prepack)
         out, _, _ = BinaryConvFunction.apply(
             x, self.weight, self.stride, self.padding, self.act_mode, t, k,
-            False)
+            False, xp, mp)
         return out
 
-    def forward_with_stats(self, x):
+    def forward_with_stats(self, x, prepack=None):
         """(out, (s1, s2)|None): per-out-channel sum/sumsq accumulated in
         the conv epilogue, for the fused BN that consumes the output.
 
@@ -235,9 +257,10 @@ class _HardBinaryConvBase(nn.Module):
         t = float(self.t) if self.t is not None else None
         k = float(self.k) if self.k is not None else None
         want = (x.is_cuda and self.training and _CONV_STATS)
+        xp, mp = self._check_prepack(x, prepack)
         out, s1, s2 = BinaryConvFunction.apply(
             x, self.weight, self.stride, self.padding, self.act_mode, t, k,
-            want)
+            want, xp, mp)
         return out, ((s1, s2) if want else None)
 
 

@@ -445,8 +445,10 @@ def test_conv_epilogue_stats_match_output_sums():
                                     True)
     ref1 = out.sum(dim=(0, 2, 3))
     ref2 = (out * out).sum(dim=(0, 2, 3))
-    assert torch.allclose(s1, ref1, rtol=1e-4, atol=1e-2)
-    assert torch.allclose(s2, ref2, rtol=1e-4, atol=1e-1)
+    # s1/s2 are [32][K] sliced partials (contention fix) — fold slices
+    assert s1.shape == (32, 96) and s2.shape == (32, 96)
+    assert torch.allclose(s1.sum(0), ref1, rtol=1e-4, atol=1e-2)
+    assert torch.allclose(s2.sum(0), ref2, rtol=1e-4, atol=1e-1)
 
 
 def test_block_with_fused_stats_matches_composition():
@@ -718,7 +720,7 @@ def test_conv_wgrad2_matches_reference(N, C, H, K):
     g = _cl(torch.randn(N, K, H, H, device="cuda", dtype=torch.bfloat16))
     xp = nat.sign_pack_nhwc(_cl(x))
     xcp = nat.repack_cplane(xp, C, H)
-    dwT = nat.conv_wgrad2(g, xcp, C)           # [9][C][K] fp32
+    dwT = nat.conv_wgrad2(g, xcp, C).sum(0)    # [nslab][9][C][K] fp32
     xb = binsign(x).to(torch.bfloat16)
     ref = torch.ops.aten.convolution_backward(
         g.float(), _cl(xb.float()),
@@ -755,11 +757,12 @@ def test_wgrad_finish_transpose_and_mask():
     torch.manual_seed(43)
     nat = _nat()
     K, C = 64, 64
-    dwT = torch.randn(9, C, K, device="cuda")
+    dwT = torch.randn(3, 9, C, K, device="cuda")   # 3 m-split slabs
     w = (torch.randn(K, C, 3, 3, device="cuda") * 1.2)
     dw = nat.wgrad_finish(dwT, w)
-    ref = dwT.permute(2, 1, 0).reshape(K, C, 3, 3) * (w.abs() <= 1).float()
-    assert torch.equal(dw, ref)
+    ref = (dwT.sum(0).permute(2, 1, 0).reshape(K, C, 3, 3)
+           * (w.abs() <= 1).float())
+    assert torch.allclose(dw, ref, atol=1e-5)
 
 
 def test_full_packed_backward_mfma_vs_miopen():
@@ -791,3 +794,60 @@ def test_full_packed_backward_mfma_vs_miopen():
             (dx2 - dx1).abs().max().item()
         assert torch.allclose(dw2, dw1, atol=5e-1, rtol=2e-2), \
             (dw2 - dw1).abs().max().item()
+
+
+# ---------------- BN-epilogue pack fusion ----------------
+
+def test_bn_epilogue_pack_matches_pack_kernel():
+    """bn_act_fwd_train(want_pack=True) bitplanes == sign_mask_pack of
+    the written output, bit-exact, both dtypes and all act kinds."""
+    nat = _nat()
+    torch.manual_seed(45)
+    for dtype in (torch.bfloat16, torch.float32):
+        for act_kind in (0, 1, 2):
+            x = _cl(torch.randn(4, 64, 14, 14, device="cuda",
+                                dtype=dtype)) * 1.5
+            gamma = torch.randn(64, device="cuda").abs() + 0.5
+            beta = torch.randn(64, device="cuda") * 0.1
+            a = torch.rand(64, device="cuda") * 0.3
+            rm = torch.zeros(64, device="cuda")
+            rv = torch.ones(64, device="cuda")
+            res = nat.bn_act_fwd_train(x, None, gamma, beta,
+                                       a if act_kind == 1 else None,
+                                       rm, rv, 0.1, 1e-5, act_kind,
+                                       None, None, True)
+            out, xpk, mpk = res[0], res[4], res[5]
+            xp_ref, mp_ref = nat.sign_mask_pack_nhwc(out)
+            assert torch.equal(xpk.flatten(), xp_ref.flatten()), \
+                (dtype, act_kind)
+            assert torch.equal(mpk.flatten(), mp_ref.flatten()), \
+                (dtype, act_kind)
+
+
+def test_resnet_block_chain_with_pack_fusion():
+    """Two chained BiBasicBlocks on GPU (pack hand-off across blocks)
+    against the CPU oracle: same loss, close grads."""
+    from bdbnn_amd.models.resnet_common import BiBasicBlock
+    torch.manual_seed(46)
+    chain_cpu = torch.nn.Sequential(BiBasicBlock(64, 64),
+                                    BiBasicBlock(64, 64))
+    chain_gpu = torch.nn.Sequential(BiBasicBlock(64, 64),
+                                    BiBasicBlock(64, 64))
+    chain_gpu.load_state_dict(chain_cpu.state_dict())
+    chain_gpu = chain_gpu.cuda().to(memory_format=torch.channels_last)
+    x = torch.randn(4, 64, 14, 14)
+    xg = _cl(x.cuda()).requires_grad_(True)
+    xc = x.clone().requires_grad_(True)
+    out_g = chain_gpu(xg)
+    if isinstance(out_g, tuple):
+        out_g = out_g[0]
+    out_c = chain_cpu(xc)
+    lg = out_g.float().pow(2).mean()
+    lc = out_c.pow(2).mean()
+    assert abs(lg.item() - lc.item()) < 5e-3, (lg.item(), lc.item())
+    lg.backward()
+    lc.backward()
+    gw_g = chain_gpu[0].conv1.weight.grad.cpu()
+    gw_c = chain_cpu[0].conv1.weight.grad
+    assert torch.allclose(gw_g, gw_c, atol=5e-3, rtol=5e-2), \
+        (gw_g - gw_c).abs().max().item()
