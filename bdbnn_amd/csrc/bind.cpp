@@ -81,7 +81,8 @@ void bdbnn_bn_act_fwd(const void*, const void*, const float*, const float*,
                       hipStream_t);
 void bdbnn_bn_act_bwd_reduce(const void*, const void*, const void*,
                              const float*, const float*, const float*,
-                             float*, int64_t, int, int, bool, hipStream_t);
+                             float*, float*, int64_t, int, int, bool,
+                             hipStream_t);
 void bdbnn_bn_act_bwd_apply(const void*, const void*, const void*,
                             const float*, const float*, const float*,
                             const float*, const float*, void*, void*,
@@ -504,8 +505,9 @@ std::vector<at::Tensor> bn_act_fwd_train(
     nslice = (int)(s1.numel() / C);
     TORCH_CHECK(nslice * C == s1.numel(), "bn: stats shape mismatch");
   } else {
-    s1 = at::empty({C}, fopt);
-    s2 = at::empty({C}, fopt);
+    s1 = at::empty({32, C}, fopt);   // sliced partials (finalize folds)
+    s2 = at::empty({32, C}, fopt);
+    nslice = 32;
     bdbnn_bn_stats(xc.data_ptr(), s1.data_ptr<float>(),
                    s2.data_ptr<float>(), n, C, bf16, cur_stream());
   }
@@ -574,6 +576,7 @@ std::vector<at::Tensor> bn_act_bwd(
   int64_t n = xc.numel();
   bool bf16 = is_bf16(xc);
   auto fopt = xc.options().dtype(at::kFloat);
+  auto sums32 = at::empty({32, C, 3}, fopt);  // sliced partials
   auto sums = at::empty({C, 3}, fopt);
   at::Tensor af;
   const float* a_ptr = nullptr;
@@ -581,7 +584,8 @@ std::vector<at::Tensor> bn_act_bwd(
                        a_ptr = af.data_ptr<float>(); }
   bdbnn_bn_act_bwd_reduce(dyc.data_ptr(), zc.data_ptr(), xc.data_ptr(),
                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                          a_ptr, sums.data_ptr<float>(), n, C, (int)act_kind,
+                          a_ptr, sums32.data_ptr<float>(),
+                          sums.data_ptr<float>(), n, C, (int)act_kind,
                           bf16, cur_stream());
   auto gf = gamma.contiguous().to(at::kFloat);
   auto dx = at::empty_like(xc, xc.options(), at::MemoryFormat::ChannelsLast);

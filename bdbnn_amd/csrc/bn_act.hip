@@ -43,9 +43,12 @@ __global__ void bn_stats_kernel(const T* __restrict__ x,
     atomicAdd(&l2[m.c0 + j], a2[j]);
   }
   __syncthreads();
+  // 32-way sliced output ([32][C], folded by bn_finalize): full-grid
+  // atomics onto C words serialize ~2048-deep otherwise
+  const int off = (blockIdx.x & 31) * C;
   for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    if (l1[c] != 0.f) atomicAdd(&s1[c], l1[c]);
-    if (l2[c] != 0.f) atomicAdd(&s2[c], l2[c]);
+    if (l1[c] != 0.f) atomicAdd(&s1[off + c], l1[c]);
+    if (l2[c] != 0.f) atomicAdd(&s2[off + c], l2[c]);
   }
 }
 
@@ -150,7 +153,10 @@ __global__ void bn_act_fwd_kernel(const T* __restrict__ x,
 }
 
 // ---- backward pass 1: per-channel reductions ----
-// sums layout: [C][3] = (sum dz, sum dz*xhat, da)
+// sums layout: [32 slices][C][3] = (sum dz, sum dz*xhat, da) partials —
+// 32-way sliced so the end-of-kernel global atomics see 1/32 of the
+// per-word contention (2048 blocks onto 3C words measured as a >20 us
+// serial tail); bn_fold_sums_kernel folds the slices.
 template <typename T, int ACT>
 __global__ void bn_act_bwd_reduce_kernel(
     const T* __restrict__ dy, const T* __restrict__ z,
@@ -201,11 +207,23 @@ __global__ void bn_act_bwd_reduce_kernel(
     if (ACT == 1) atomicAdd(&r2[m.c0 + j], s2[j]);
   }
   __syncthreads();
+  float* my = sums + (blockIdx.x & 31) * (C * 3);
   for (int cc = threadIdx.x; cc < C; cc += blockDim.x) {
-    if (r0[cc] != 0.f) atomicAdd(&sums[cc * 3 + 0], r0[cc]);
-    if (r1[cc] != 0.f) atomicAdd(&sums[cc * 3 + 1], r1[cc]);
-    if (r2[cc] != 0.f) atomicAdd(&sums[cc * 3 + 2], r2[cc]);
+    if (r0[cc] != 0.f) atomicAdd(&my[cc * 3 + 0], r0[cc]);
+    if (r1[cc] != 0.f) atomicAdd(&my[cc * 3 + 1], r1[cc]);
+    if (r2[cc] != 0.f) atomicAdd(&my[cc * 3 + 2], r2[cc]);
   }
+}
+
+// fold the 32 slices -> [C][3] (tiny; launched right after the reduce)
+__global__ void bn_fold_sums_kernel(const float* __restrict__ sums32,
+                                    float* __restrict__ out, int n) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  float v = 0.f;
+#pragma unroll
+  for (int s = 0; s < 32; ++s) v += sums32[s * n + i];
+  out[i] = v;
 }
 
 // ---- backward pass 2: dx (+ dskip) ----
@@ -371,8 +389,8 @@ static void launch_eval(const void* x, const void* skip, const float* rm,
 extern "C" void bdbnn_bn_stats(const void* x, float* s1, float* s2,
                                int64_t n, int C, bool bf16,
                                hipStream_t stream) {
-  hipMemsetAsync(s1, 0, sizeof(float) * C, stream);
-  hipMemsetAsync(s2, 0, sizeof(float) * C, stream);
+  hipMemsetAsync(s1, 0, sizeof(float) * 32 * C, stream);
+  hipMemsetAsync(s2, 0, sizeof(float) * 32 * C, stream);
   int64_t n_pix = n / C;
   int grid = grid_pix8(n_pix, C);
   size_t lds = 2 * sizeof(float) * C;
@@ -414,19 +432,21 @@ extern "C" void bdbnn_bn_act_fwd(const void* x, const void* skip,
 extern "C" void bdbnn_bn_act_bwd_reduce(const void* dy, const void* z,
                                         const void* x, const float* mean,
                                         const float* invstd, const float* a,
-                                        float* sums, int64_t n, int C,
-                                        int act_kind, bool bf16,
-                                        hipStream_t stream) {
-  hipMemsetAsync(sums, 0, sizeof(float) * C * 3, stream);
+                                        float* sums32, float* sums,
+                                        int64_t n, int C, int act_kind,
+                                        bool bf16, hipStream_t stream) {
+  hipMemsetAsync(sums32, 0, sizeof(float) * 32 * C * 3, stream);
   int64_t n_pix = n / C;
   int grid = grid_pix8(n_pix, C);
   size_t lds = 3 * sizeof(float) * C;
   if (bf16)
-    launch_bwd_reduce<uint16_t>(dy, z, x, mean, invstd, a, sums, n_pix, C,
+    launch_bwd_reduce<uint16_t>(dy, z, x, mean, invstd, a, sums32, n_pix, C,
                                 act_kind, grid, lds, stream);
   else
-    launch_bwd_reduce<float>(dy, z, x, mean, invstd, a, sums, n_pix, C,
+    launch_bwd_reduce<float>(dy, z, x, mean, invstd, a, sums32, n_pix, C,
                              act_kind, grid, lds, stream);
+  bn_fold_sums_kernel<<<(C * 3 + 255) / 256, 256, 0, stream>>>(sums32, sums,
+                                                               C * 3);
 }
 
 extern "C" void bdbnn_bn_act_bwd_apply(const void* dy, const void* z,
