@@ -31,11 +31,27 @@ __global__ void bn_stats_kernel(const T* __restrict__ x,
   __syncthreads();
   ChanMap m = chan_map8(C);
   float a1[8] = {}, a2[8] = {};
-  float v[8];
-  for (int64_t p = m.p0; p < n_pix; p += m.pstep) {
-    load8(x, p * C + m.c0, v);
+  // 4 independent pixel loads in flight per wave: one blocking load per
+  // iteration leaves the wave latency-bound at ~0.6 B/cyc/CU (34 VGPR,
+  // plenty of headroom for the extra 48 registers)
+  float v0[8], v1[8], v2[8], v3[8];
+  int64_t p = m.p0;
+  for (; p + 3 * m.pstep < n_pix; p += 4 * m.pstep) {
+    load8(x, p * C + m.c0, v0);
+    load8(x, (p + m.pstep) * C + m.c0, v1);
+    load8(x, (p + 2 * m.pstep) * C + m.c0, v2);
+    load8(x, (p + 3 * m.pstep) * C + m.c0, v3);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) { a1[j] += v[j]; a2[j] += v[j] * v[j]; }
+    for (int j = 0; j < 8; ++j) {
+      a1[j] += v0[j] + v1[j] + v2[j] + v3[j];
+      a2[j] += v0[j] * v0[j] + v1[j] * v1[j] + v2[j] * v2[j] +
+               v3[j] * v3[j];
+    }
+  }
+  for (; p < n_pix; p += m.pstep) {
+    load8(x, p * C + m.c0, v0);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { a1[j] += v0[j]; a2[j] += v0[j] * v0[j]; }
   }
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
@@ -180,26 +196,49 @@ __global__ void bn_act_bwd_reduce_kernel(
     av[j] = (ACT == 1) ? a[cc] : 0.f;
   }
   float s0[8] = {}, s1[8] = {}, s2[8] = {};
-  float dyv[8], zv[8], xv[8];
-  for (int64_t p = m.p0; p < n_pix; p += m.pstep) {
+  // 2 pixel iterations in flight (6 independent loads): one blocking
+  // triple per iteration leaves the wave latency-bound.  (The r1
+  // attempt unrolled with a COMBINED body and regressed on register
+  // pressure; issuing the loads up front then reducing keeps liveness
+  // to the 6 vector buffers.)
+  float dyv[8], zv[8], xv[8], dyw[8], zw[8], xw[8];
+#define BN_RED_BODY(DYV, ZV, XV)                                          \
+  {                                                                       \
+    _Pragma("unroll")                                                     \
+    for (int j = 0; j < 8; ++j) {                                         \
+      float dz = DYV[j];                                                  \
+      if (ACT == 1) {                                                     \
+        dz = ZV[j] > 0.f ? DYV[j] : av[j] * DYV[j];                       \
+        if (ZV[j] <= 0.f) s2[j] += DYV[j] * ZV[j];                        \
+      } else if (ACT == 2) {                                              \
+        dz = ZV[j] > 0.f ? DYV[j] : 0.f;                                  \
+      }                                                                   \
+      float xhat = (XV[j] - mu[j]) * is[j];                               \
+      s0[j] += dz;                                                        \
+      s1[j] += dz * xhat;                                                 \
+    }                                                                     \
+  }
+  int64_t p = m.p0;
+  for (; p + m.pstep < n_pix; p += 2 * m.pstep) {
+    int64_t i = p * C + m.c0;
+    int64_t i2 = (p + m.pstep) * C + m.c0;
+    load8(dy, i, dyv);
+    load8(x, i, xv);
+    if (ACT != 0) load8(z, i, zv);
+    load8(dy, i2, dyw);
+    load8(x, i2, xw);
+    if (ACT != 0) load8(z, i2, zw);
+    BN_RED_BODY(dyv, zv, xv)
+    BN_RED_BODY(dyw, zw, xw)
+  }
+  for (; p < n_pix; p += m.pstep) {
     int64_t i = p * C + m.c0;
     load8(dy, i, dyv);
     load8(x, i, xv);
     if (ACT != 0) load8(z, i, zv);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      float dz = dyv[j];
-      if (ACT == 1) {
-        dz = zv[j] > 0.f ? dyv[j] : av[j] * dyv[j];
-        if (zv[j] <= 0.f) s2[j] += dyv[j] * zv[j];
-      } else if (ACT == 2) {
-        dz = zv[j] > 0.f ? dyv[j] : 0.f;
-      }
-      float xhat = (xv[j] - mu[j]) * is[j];
-      s0[j] += dz;
-      s1[j] += dz * xhat;
-    }
+    BN_RED_BODY(dyv, zv, xv)
   }
+#undef BN_RED_BODY
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     atomicAdd(&r0[m.c0 + j], s0[j]);
