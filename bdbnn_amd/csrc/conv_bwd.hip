@@ -435,7 +435,7 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
 
   // X: [3 copies][64 c] rows of GB*XROWS*WP (+pad) elements
   __shared__ __align__(16) __bf16 X[3 * WG2_BC * XSTRIDE];
-  __shared__ __align__(16) __bf16 gT[WG2_BK * GSTRIDE];
+  __shared__ __align__(16) __bf16 gT[2][WG2_BK * GSTRIDE];
   __shared__ __align__(16) __bf16 lut[256][8];
   __shared__ float red[4][32][32];        // m-split combine scratch
   // raw xcp row bits: the global u64 loads prefetch into registers
@@ -471,47 +471,41 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
   int ch = ch_lo;
 
   // staging assignments (fixed per thread)
-  //   gT: thread loads 8 k-elements of TWO adjacent m rows and writes 8
-  //   paired b32 stores (half the LDS-write instructions of a b16
-  //   scatter transpose).
-  const int sg_m2 = (tid & 63) * 2;       // m pair base
-  const int sg_k8 = (tid >> 6) * 8;       // its 8 k elements
+  //   gT: thread loads 16 k-elements of ONE pixel (32 B contiguous; a
+  //   4-lane group covers one pixel's 64-ch row = 128 B coalesced) and
+  //   scatter-writes them transposed.  (A 2-pixel/8-ch variant halved
+  //   the LDS stores but scattered the GLOBAL loads to 64 cache lines
+  //   per wave — measured 30-60% slower on the WP<=16 classes.)
+  const int sg_m = tid >> 2;              // 0..127
+  const int sg_k16 = (tid & 3) * 16;      // k offset of its 16 elements
   uint4 greg[2];
-
-#define G_ADDR(chunk, m, out_ptr)                                         \
-  {                                                                       \
-    int band = (m) / (RB * WP);                                           \
-    int rem = (m)-band * (RB * WP);                                       \
-    int slot = (chunk)*GB + band;                                         \
-    int x = rem & (WP - 1);                                               \
-    out_ptr = nullptr;                                                    \
-    if (slot < p.total_slots && x < p.W) {                                \
-      int n = slot / p.bands_per_image;                                   \
-      int y = (slot - n * p.bands_per_image) * RB + (rem >> LGWP);        \
-      if (y < p.H)                                                        \
-        out_ptr = g + ((int64_t)(n * p.H + y) * p.W + x) * p.K + k0 +     \
-                  sg_k8;                                                  \
-    }                                                                     \
-  }
 
 #define G_LOAD(chunk)                                                     \
   {                                                                       \
-    const __bf16* src0;                                                   \
-    const __bf16* src1;                                                   \
-    G_ADDR(chunk, sg_m2, src0);                                           \
-    G_ADDR(chunk, sg_m2 + 1, src1);                                       \
+    int band = sg_m / (RB * WP);                                          \
+    int rem = sg_m - band * (RB * WP);                                    \
+    int slot = (chunk)*GB + band;                                         \
+    int x = rem & (WP - 1);                                               \
     uint4 z{0, 0, 0, 0};                                                  \
-    greg[0] = src0 ? *(const uint4*)src0 : z;                             \
-    greg[1] = src1 ? *(const uint4*)src1 : z;                             \
+    greg[0] = z; greg[1] = z;                                             \
+    if (slot < p.total_slots && x < p.W) {                                \
+      int n = slot / p.bands_per_image;                                   \
+      int y = (slot - n * p.bands_per_image) * RB + (rem >> LGWP);        \
+      if (y < p.H) {                                                      \
+        const __bf16* src =                                               \
+            g + ((int64_t)(n * p.H + y) * p.W + x) * p.K + k0 + sg_k16;   \
+        greg[0] = *(const uint4*)src;                                     \
+        greg[1] = *(const uint4*)(src + 8);                               \
+      }                                                                   \
+    }                                                                     \
   }
 
-#define G_WRITE()                                                         \
+#define G_WRITE(buf)                                                      \
   {                                                                       \
     _Pragma("unroll")                                                     \
-    for (int j = 0; j < 8; ++j) {                                         \
-      uint32_t pair = (uint32_t)((const uint16_t*)greg)[j] |              \
-                      ((uint32_t)((const uint16_t*)&greg[1])[j] << 16);   \
-      *(uint32_t*)&gT[(sg_k8 + j) * GSTRIDE + sg_m2] = pair;              \
+    for (int j = 0; j < 16; ++j) {                                        \
+      __bf16 v = ((const __bf16*)greg)[j];                                \
+      gT[buf][(sg_k16 + j) * GSTRIDE + sg_m] = v;                         \
     }                                                                     \
   }
 
@@ -627,8 +621,9 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
     BITS_LOAD(ch, 0, XROWS);
     G_LOAD(ch);
     BITS_WRITE(ch, 0, XROWS);
-    G_WRITE();
+    G_WRITE(0);
   }
+  int gb = 0;
 
   // per-lane fragment bases
   // A (X): lane row c = qc*32 + lrow; m-run start offset = lhalf*8
@@ -660,7 +655,8 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
     for (int msl = 0; msl < 4; ++msl) {   // 4 x 16-m steps = its 64-m half
       const int m16 = ms_grp * 64 + msl * 16;
       // B-frag: 8 m at fixed k from gT
-      bf16x8 bfrag = *(const bf16x8*)&gT[b_k * GSTRIDE + m16 + lhalf * 8];
+      bf16x8 bfrag =
+          *(const bf16x8*)&gT[gb][b_k * GSTRIDE + m16 + lhalf * 8];
       // A-frags per tap from X: ring plane (rloc + dy), x-run
       const int mstart = m16 + lhalf * 8;
       const int rl = mstart >> LGWP;
@@ -684,12 +680,13 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
                                                     0, 0, 0);
       }
     }
-    __syncthreads();                  // all reads of X/gT done
+    __syncthreads();                  // all reads of X/gT[gb] done
     if (more) {
       if (cont_n) BITS_WRITE(ch + 1, 2, RB)
       else BITS_WRITE(ch + 1, 0, XROWS)
-      G_WRITE();
+      G_WRITE(gb ^ 1);
     }
+    gb ^= 1;
     ++ch;
   }
 

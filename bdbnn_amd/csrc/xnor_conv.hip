@@ -196,6 +196,7 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
     int kg = k0_blk + kq + j;
     al[j] = (kg < p.K) ? alpha[kg] : 0.f;
   }
+  float st1[STATS ? 4 : 1] = {}, st2[STATS ? 4 : 1] = {};
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
     int64_t sp = m0 + r0 + i;
@@ -227,8 +228,7 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
           float v = bf16_to_f32(vals[j]);   // stats of the ROUNDED value
-          atomicAdd(&csum[0][kq + j], v);
-          atomicAdd(&csum[1][kq + j], v * v);
+          st1[j] += v; st2[j] += v * v;
         }
     } else {
       float vals[4];
@@ -244,13 +244,18 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
       if constexpr (STATS)
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
-          atomicAdd(&csum[0][kq + j], vals[j]);
-          atomicAdd(&csum[1][kq + j], vals[j] * vals[j]);
+          st1[j] += vals[j]; st2[j] += vals[j] * vals[j];
         }
     }
   }
   if constexpr (STATS) {
-    __syncthreads();  // per-row LDS adds above visible block-wide
+    __syncthreads();  // csum init visible
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      atomicAdd(&csum[0][kq + j], st1[j]);
+      atomicAdd(&csum[1][kq + j], st2[j]);
+    }
+    __syncthreads();
     // 32-way sliced accumulators ([32][K], summed by bn_finalize): with
     // one flat [K] buffer ~12k blocks contend on K words and the whole
     // kernel slows ~15% e2e (r1 optimization log); 32 slices cut the

@@ -31,7 +31,7 @@ import torch.nn.functional as F
 from .binarize import binsign, weight_scale
 from .. import _C
 
-_CONV_STATS = os.environ.get("BDBNN_CONV_STATS", "1") != "0"
+_CONV_STATS = os.environ.get("BDBNN_CONV_STATS", "0") == "1"
 # hand-written MFMA backward kernels (default ON; BDBNN_MFMA_BWD=0 falls
 # back to MIOpen igemm on decoded operands for A/B comparison)
 _MFMA_BWD = os.environ.get("BDBNN_MFMA_BWD", "1") != "0"
@@ -249,11 +249,12 @@ class _HardBinaryConvBase(nn.Module):
         """(out, (s1, s2)|None): per-out-channel sum/sumsq accumulated in
         the conv epilogue, for the fused BN that consumes the output.
 
-        ON by default since the epilogue went to 32-way sliced
-        accumulators ([32][K] partials, folded by bn_finalize): the r1
-        flat-[K] version lost 15% e2e to atomic contention with ~12k
-        blocks hitting K words.  BDBNN_CONV_STATS=0 disables (BN then
-        runs its own stats read pass) for A/B."""
+        OFF by default, MEASURED twice (r1 flat-[K] atomics, r2 32-way
+        sliced + LDS accumulation): any stats work in this epilogue
+        costs the conv kernel 1.4-4x (profiles/, bn_bench.md r2) —
+        far more than the separate bn_stats read pass it saves, which
+        after the 4-deep MLP unroll costs ~0.07 ms/layer at b512.
+        BDBNN_CONV_STATS=1 re-enables for experiments."""
         t = float(self.t) if self.t is not None else None
         k = float(self.k) if self.k is not None else None
         want = (x.is_cuda and self.training and _CONV_STATS)
