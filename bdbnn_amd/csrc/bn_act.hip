@@ -285,24 +285,45 @@ __global__ void bn_act_bwd_apply_kernel(
     sdz_n[j] = sums[cc * 3 + 0] * inv_n;
     sdzx_n[j] = sums[cc * 3 + 1] * inv_n;
   }
+  // 2 pixel iterations in flight (up to 6 independent loads): one
+  // blocking triple per iteration leaves the wave latency-bound
   float dyv[8], zv[8], xv[8], dxv[8], dzv[8];
-  for (int64_t p = m.p0; p < n_pix; p += m.pstep) {
+  float dyw[8], zw[8], xw[8];
+#define BN_APPLY_BODY(DYV, ZV, XV, I)                                     \
+  {                                                                       \
+    _Pragma("unroll")                                                     \
+    for (int j = 0; j < 8; ++j) {                                         \
+      float dz = DYV[j];                                                  \
+      if (ACT == 1) dz = ZV[j] > 0.f ? DYV[j] : av[j] * DYV[j];           \
+      else if (ACT == 2) dz = ZV[j] > 0.f ? DYV[j] : 0.f;                 \
+      float xhat = (XV[j] - mu[j]) * is[j];                               \
+      dxv[j] = gis[j] * (dz - sdz_n[j] - xhat * sdzx_n[j]);               \
+      dzv[j] = dz;                                                        \
+    }                                                                     \
+    store8(dx, I, dxv);                                                   \
+    if (dskip != nullptr) store8(dskip, I, dzv);                          \
+  }
+  int64_t p = m.p0;
+  for (; p + m.pstep < n_pix; p += 2 * m.pstep) {
+    int64_t i = p * C + m.c0;
+    int64_t i2 = (p + m.pstep) * C + m.c0;
+    load8(dy, i, dyv);
+    load8(x, i, xv);
+    if (ACT != 0) load8(z, i, zv);
+    load8(dy, i2, dyw);
+    load8(x, i2, xw);
+    if (ACT != 0) load8(z, i2, zw);
+    BN_APPLY_BODY(dyv, zv, xv, i)
+    BN_APPLY_BODY(dyw, zw, xw, i2)
+  }
+  for (; p < n_pix; p += m.pstep) {
     int64_t i = p * C + m.c0;
     load8(dy, i, dyv);
     load8(x, i, xv);
     if (ACT != 0) load8(z, i, zv);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      float dz = dyv[j];
-      if (ACT == 1) dz = zv[j] > 0.f ? dyv[j] : av[j] * dyv[j];
-      else if (ACT == 2) dz = zv[j] > 0.f ? dyv[j] : 0.f;
-      float xhat = (xv[j] - mu[j]) * is[j];
-      dxv[j] = gis[j] * (dz - sdz_n[j] - xhat * sdzx_n[j]);
-      dzv[j] = dz;
-    }
-    store8(dx, i, dxv);
-    if (dskip != nullptr) store8(dskip, i, dzv);
+    BN_APPLY_BODY(dyv, zv, xv, i)
   }
+#undef BN_APPLY_BODY
 }
 
 // ---- eval-mode fused normalize(+add)(+act) using running stats ----
