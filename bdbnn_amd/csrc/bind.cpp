@@ -61,6 +61,14 @@ void bdbnn_dgrad_wdec(const uint32_t*, const float*, void*, int, int,
 int bdbnn_conv_wgrad2(const void*, const uint64_t*, float*, int, int, int,
                       int, int, hipStream_t);
 int bdbnn_wgrad2_nslab(int, int, int, int, int);
+void bdbnn_stem_pack_x4(const void*, void*, int64_t, bool, hipStream_t);
+void bdbnn_stem_pack_w4(const float*, void*, hipStream_t);
+void bdbnn_stem_fwd(const void*, const void*, void*, int, int, int,
+                    hipStream_t);
+int bdbnn_stem_wrw_nslab(int, int, int);
+void bdbnn_stem_wrw(const void*, const void*, float*, int, int, int,
+                    hipStream_t);
+void bdbnn_stem_fold_dw4(const float*, float*, int, hipStream_t);
 void bdbnn_repack_cplane(const uint32_t*, uint64_t*, int, int, int, int,
                          hipStream_t);
 void bdbnn_wgrad_finish(const float*, const float*, float*, int, int, int,
@@ -355,6 +363,53 @@ std::vector<at::Tensor> xnor_conv_fwd(
                       cur_stream());
   if (want_stats) return {out, s1, s2};
   return {out};
+}
+
+
+// ---------------- stem conv (7x7/2, C=3 -> 64) ----------------
+
+std::vector<at::Tensor> stem_conv_fwd(const at::Tensor& x,
+                                      const at::Tensor& w) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.size(1) == 3,
+              "stem: (N,3,H,W) input");
+  TORCH_CHECK(w.dim() == 4 && w.size(0) == 64 && w.size(1) == 3 &&
+                  w.size(2) == 7 && w.size(3) == 7,
+              "stem: (64,3,7,7) weights");
+  int N = (int)x.size(0), H = (int)x.size(2), W = (int)x.size(3);
+  TORCH_CHECK(H % 2 == 0 && W % 2 == 0, "stem: even H,W");
+  auto xc = x.contiguous(at::MemoryFormat::ChannelsLast);
+  bool bf16 = is_bf16(xc);
+  auto bopt = xc.options().dtype(at::kBFloat16);
+  auto x4 = at::empty({N, H, W, 4}, bopt);
+  bdbnn_stem_pack_x4(xc.data_ptr(), x4.data_ptr(), (int64_t)N * H * W,
+                     bf16, cur_stream());
+  auto wf = w.contiguous().to(at::kFloat);
+  auto w4T = at::empty({64, 224}, bopt);
+  bdbnn_stem_pack_w4(wf.data_ptr<float>(), w4T.data_ptr(), cur_stream());
+  auto out = at::empty({N, 64, H / 2, W / 2}, bopt,
+                       at::MemoryFormat::ChannelsLast);
+  bdbnn_stem_fwd(x4.data_ptr(), w4T.data_ptr(), out.data_ptr(), N, H, W,
+                 cur_stream());
+  return {out, x4};
+}
+
+at::Tensor stem_conv_wrw(const at::Tensor& x4, const at::Tensor& gy) {
+  TORCH_CHECK(x4.is_cuda() && x4.dim() == 4 && x4.size(3) == 4 &&
+                  x4.scalar_type() == at::kBFloat16,
+              "stem wrw: packed x4 (N,H,W,4) bf16");
+  int N = (int)x4.size(0), H = (int)x4.size(1), W = (int)x4.size(2);
+  auto gc = gy.to(at::kBFloat16).contiguous(at::MemoryFormat::ChannelsLast);
+  TORCH_CHECK(gc.size(1) == 64 && gc.size(2) == H / 2 &&
+                  gc.size(3) == W / 2, "stem wrw: grad shape");
+  int nslab = bdbnn_stem_wrw_nslab(N, H, W);
+  auto fopt = x4.options().dtype(at::kFloat);
+  auto slabs = at::empty({nslab, 64, 224}, fopt);
+  bdbnn_stem_wrw(x4.data_ptr(), gc.data_ptr(), slabs.data_ptr<float>(),
+                 N, H, W, cur_stream());
+  auto dw = at::empty({64, 3, 7, 7}, fopt);
+  bdbnn_stem_fold_dw4(slabs.data_ptr<float>(), dw.data_ptr<float>(),
+                      nslab, cur_stream());
+  return dw;
 }
 
 // ---------------- fused classification losses ----------------
@@ -914,6 +969,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mask_mul_packed", &mask_mul_packed, "dx = mask_bit ? g : 0");
   m.def("weight_decode", &weight_decode, "packed weights -> alpha*(+-1)");
   m.def("xnor_conv_fwd", &xnor_conv_fwd, "bit-packed XNOR+popcount conv");
+  m.def("stem_conv_fwd", &stem_conv_fwd,
+        "MFMA 7x7/2 stem conv fwd (returns out, packed x4)");
+  m.def("stem_conv_wrw", &stem_conv_wrw, "MFMA stem conv weight grad");
   m.def("kd_logit_fwd", &kd_logit_fwd, "fused logit-KD fwd");
   m.def("kd_logit_bwd", &kd_logit_bwd, "fused logit-KD bwd");
   m.def("ce_fwd", &ce_fwd, "fused cross-entropy fwd");

@@ -480,20 +480,20 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
   const int sg_k16 = (tid & 3) * 16;      // k offset of its 16 elements
   uint4 greg[2];
 
+// (uses the CHUNK_INFO of the same chunk — call order guarantees it)
 #define G_LOAD(chunk)                                                     \
   {                                                                       \
     int band = sg_m / (RB * WP);                                          \
     int rem = sg_m - band * (RB * WP);                                    \
-    int slot = (chunk)*GB + band;                                         \
     int x = rem & (WP - 1);                                               \
     uint4 z{0, 0, 0, 0};                                                  \
     greg[0] = z; greg[1] = z;                                             \
-    if (slot < p.total_slots && x < p.W) {                                \
-      int n = slot / p.bands_per_image;                                   \
-      int y = (slot - n * p.bands_per_image) * RB + (rem >> LGWP);        \
+    if (u_vs[band] && x < p.W) {                                          \
+      int y = u_y0[band] + (rem >> LGWP);                                 \
       if (y < p.H) {                                                      \
         const __bf16* src =                                               \
-            g + ((int64_t)(n * p.H + y) * p.W + x) * p.K + k0 + sg_k16;   \
+            g + ((int64_t)(u_n[band] * p.H + y) * p.W + x) * p.K + k0 +   \
+            sg_k16;                                                       \
         greg[0] = *(const uint4*)src;                                     \
         greg[1] = *(const uint4*)(src + 8);                               \
       }                                                                   \
@@ -518,6 +518,22 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
   uint64_t bits_reg[BPT];
   unsigned char valid_reg[BPT];
   int nbits_cur = 0;
+  // block-uniform per-band facts of a chunk (one divide per band, not
+  // one per decode target)
+  int u_n[GB], u_y0[GB], u_pb[GB];
+  bool u_vs[GB];
+#define CHUNK_INFO(chunk)                                                 \
+  {                                                                       \
+    _Pragma("unroll")                                                     \
+    for (int b = 0; b < GB; ++b) {                                        \
+      int slot = (chunk)*GB + b;                                          \
+      u_vs[b] = slot < p.total_slots;                                     \
+      int n = slot / p.bands_per_image;                                   \
+      u_n[b] = n;                                                         \
+      u_y0[b] = (slot - n * p.bands_per_image) * RB;                      \
+      u_pb[b] = RING ? u_y0[b] % XROWS : 0;                               \
+    }                                                                     \
+  }
 #define BITS_LOAD(chunk, r0, NR)                                          \
   {                                                                       \
     nbits_cur = WG2_BC * GB * (NR);                                       \
@@ -531,12 +547,10 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
         int rem = i / (NR);                                               \
         int band = rem % GB;                                              \
         int c = rem / GB;                                                 \
-        int slot = (chunk)*GB + band;                                     \
-        if (slot < p.total_slots) {                                       \
-          int n = slot / p.bands_per_image;                               \
-          int y = (slot - n * p.bands_per_image) * RB + w - 1;            \
+        if (u_vs[band]) {                                                 \
+          int y = u_y0[band] + w - 1;                                     \
           if (y >= 0 && y < p.H) {                                        \
-            b = xcp[(int64_t)(c0 + c) * p.nh_rows + n * p.H + y];         \
+            b = xcp[(int64_t)(c0 + c) * p.nh_rows + u_n[band] * p.H + y]; \
             v = 1;                                                        \
           }                                                               \
         }                                                                 \
@@ -555,13 +569,8 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
         int rem = i / (NR);                                               \
         int band = rem % GB;                                              \
         int c = rem / GB;                                                 \
-        int plane = w;                                                    \
-        if (RING) {                                                       \
-          int slot = (chunk)*GB + band;                                   \
-          int y0 = (slot - (slot / p.bands_per_image) *                   \
-                              p.bands_per_image) * RB;                    \
-          plane = (y0 + w) % XROWS;                                       \
-        }                                                                 \
+        int plane = RING ? u_pb[band] + w : w;                            \
+        if (RING && plane >= XROWS) plane -= XROWS;                       \
         int bi = (c * GB + band) * XROWS + plane;                         \
         xbits[bi] = bits_reg[it];                                         \
         xvalid[bi] = valid_reg[it];                                       \
@@ -588,13 +597,8 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
         rem /= GB;                                                        \
         int c = rem % WG2_BC;                                             \
         int dxs = rem / WG2_BC;                                           \
-        int plane = w;                                                    \
-        if (RING) {                                                       \
-          int slot = (chunk)*GB + band;                                   \
-          int y0 = (slot - (slot / p.bands_per_image) *                   \
-                              p.bands_per_image) * RB;                    \
-          plane = (y0 + w) % XROWS;                                       \
-        }                                                                 \
+        int plane = RING ? u_pb[band] + w : w;                            \
+        if (RING && plane >= XROWS) plane -= XROWS;                       \
         int bi = (c * GB + band) * XROWS + plane;                         \
         uint64_t bits = xbits[bi];                                        \
         /* kw shift: X_dxs[x] = xb[x + dxs - 1] */                        \
@@ -618,6 +622,7 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
   (RING && (chunk) > ch_lo && ((chunk) % p.bands_per_image) != 0)
 
   if (n_chunks > 0) {
+    CHUNK_INFO(ch);
     BITS_LOAD(ch, 0, XROWS);
     G_LOAD(ch);
     BITS_WRITE(ch, 0, XROWS);
@@ -635,16 +640,14 @@ __global__ __launch_bounds__(512, 2) void conv_wgrad2_kernel(
     const bool more = ci + 1 < n_chunks;
     const bool cont = IS_CONT(ch);
     const bool cont_n = IS_CONT(ch + 1);
+    CHUNK_INFO(ch);
     // y0 of this chunk modulo the ring (0 when !RING: y0 == 0 there)
-    int y0m = 0;
-    if (RING) {
-      int y0 = (ch - (ch / p.bands_per_image) * p.bands_per_image) * RB;
-      y0m = y0 % XROWS;
-    }
+    const int y0m = RING ? u_pb[0] : 0;
     __syncthreads();                  // xbits + gT visible; X free
     if (cont) X_DECODE(ch, 2, RB)
     else X_DECODE(ch, 0, XROWS)
     if (more) {
+      CHUNK_INFO(ch + 1);             // (also serves BITS_WRITE below)
       if (cont_n) BITS_LOAD(ch + 1, 2, RB)
       else BITS_LOAD(ch + 1, 0, XROWS)
       G_LOAD(ch + 1);                 // lands during the MFMA phase

@@ -21,6 +21,7 @@ import torch
 import torch.nn as nn
 
 from ..ops.binary_conv import HardBinaryConv
+from ..ops.stem_conv import StemConv7x7
 from ..ops.binarize import LearnableBias
 from ..ops.activations import ChannelPReLU
 from ..ops.bn_act import fused_bn_act
@@ -137,7 +138,8 @@ class ResNet(nn.Module):
         self.binary = binary
         self.inplanes = width
         if stem == "imagenet":
-            self.conv1 = nn.Conv2d(3, width, 7, 2, 3, bias=False)
+            conv_t = StemConv7x7 if width == 64 else nn.Conv2d
+            self.conv1 = conv_t(3, width, 7, 2, 3, bias=False)
         else:
             self.conv1 = nn.Conv2d(3, width, 3, 1, 1, bias=False)
         self.bn1 = nn.BatchNorm2d(width)

@@ -1,0 +1,62 @@
+"""Hand-written MFMA stem convolution (7x7 / stride 2 / pad 3, 3 -> 64).
+
+MIOpen's igemm puts C=3 on the GEMM K-dim and runs ~15x off the MFMA
+floor (measured 2.9 ms fwd + 2.8 ms wrw per b2048 step, the largest
+non-owned block of the profile).  csrc/stem_conv.hip instead makes the
+whole 7x7x(3->4-padded) patch the K-dim (224 slots) so every fragment
+is an aligned LDS read.  See SURVEY.md K7 (stem was "acceptable" on
+MIOpen; owning it is worth ~4 ms/step at b2048).
+
+``StemConv7x7`` subclasses nn.Conv2d so state_dict / init / repr are
+untouched; anything off the fast path falls back to the stock conv.
+The stem input never requires grad in this framework, so backward only
+produces the weight gradient (loud error otherwise).
+"""
+
+import torch
+import torch.nn as nn
+
+from .. import _C
+
+
+class _StemConvFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w):
+        nat = _C.native_required()
+        out, x4 = nat.stem_conv_fwd(x, w)
+        if ctx.needs_input_grad[0]:
+            raise RuntimeError("StemConv7x7: input gradients unsupported "
+                               "(the stem input is the data batch)")
+        if ctx.needs_input_grad[1]:
+            ctx.save_for_backward(x4)
+        return out
+
+    @staticmethod
+    def backward(ctx, gy):
+        (x4,) = ctx.saved_tensors
+        nat = _C.native_required()
+        dw = nat.stem_conv_wrw(x4, gy)
+        return None, dw
+
+
+class StemConv7x7(nn.Conv2d):
+    """Drop-in for the ImageNet ResNet stem nn.Conv2d(3, 64, 7, 2, 3)."""
+
+    def forward(self, x):
+        fast = (x.is_cuda and _C.has_native()
+                and self.in_channels == 3 and self.out_channels == 64
+                and self.kernel_size == (7, 7) and self.stride == (2, 2)
+                and self.padding == (3, 3) and self.bias is None
+                and x.dim() == 4 and x.size(1) == 3
+                and x.size(2) % 2 == 0 and x.size(3) % 2 == 0
+                and not x.requires_grad
+                and x.dtype in (torch.float32, torch.bfloat16))
+        if fast:
+            out = _StemConvFn.apply(x, self.weight)
+            # autocast parity: the stock conv would emit bf16 under
+            # autocast and fp32 otherwise; our kernel is bf16-native
+            if (not torch.is_autocast_enabled()
+                    and x.dtype == torch.float32):
+                out = out.float()
+            return out
+        return super().forward(x)

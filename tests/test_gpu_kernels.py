@@ -853,3 +853,65 @@ def test_resnet_block_chain_with_pack_fusion():
     gw_c = chain_cpu[0].conv1.weight.grad
     assert torch.allclose(gw_g, gw_c, atol=5e-3, rtol=5e-2), \
         (gw_g - gw_c).abs().max().item()
+
+
+# ---------------- MFMA stem conv (7x7/2, 3->64) ----------------
+
+def test_stem_conv_fwd_matches_reference():
+    nat = _nat()
+    torch.manual_seed(47)
+    for (N, H) in [(2, 224), (3, 56)]:
+        x = _cl(torch.randn(N, 3, H, H, device="cuda",
+                            dtype=torch.bfloat16))
+        w = torch.randn(64, 3, 7, 7, device="cuda") * 0.1
+        out, x4 = nat.stem_conv_fwd(x, w)
+        assert x4.shape == (N, H, H, 4)
+        ref = torch.nn.functional.conv2d(
+            x.float(), w, None, stride=2, padding=3)
+        assert out.shape == ref.shape
+        err = (out.float() - _cl(ref)).abs().max().item()
+        assert torch.allclose(out.float(), _cl(ref), atol=0.1,
+                              rtol=2e-2), (H, err)
+
+
+def test_stem_conv_wrw_matches_reference():
+    nat = _nat()
+    torch.manual_seed(48)
+    for (N, H) in [(2, 224), (3, 56)]:
+        x = _cl(torch.randn(N, 3, H, H, device="cuda",
+                            dtype=torch.bfloat16))
+        w = torch.randn(64, 3, 7, 7, device="cuda") * 0.1
+        _, x4 = nat.stem_conv_fwd(x, w)
+        g = _cl(torch.randn(N, 64, H // 2, H // 2, device="cuda",
+                            dtype=torch.bfloat16))
+        dw = nat.stem_conv_wrw(x4, g)
+        ref = torch.ops.aten.convolution_backward(
+            g.float(), x.float(), torch.empty(64, 3, 7, 7, device="cuda"),
+            None, [2, 2], [3, 3], [1, 1], False, [0, 0], 1,
+            [False, True, False])[1]
+        err = (dw - ref).abs().max().item()
+        rel = err / ref.abs().max().item()
+        assert rel < 2e-2, (H, err, rel)
+
+
+def test_stem_module_autograd():
+    """StemConv7x7 module: fast path out + dw vs the stock conv path."""
+    from bdbnn_amd.ops.stem_conv import StemConv7x7
+    torch.manual_seed(49)
+    m = StemConv7x7(3, 64, 7, 2, 3, bias=False).cuda()
+    m = m.to(memory_format=torch.channels_last)
+    x = _cl(torch.randn(4, 3, 64, 64, device="cuda"))
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        out = m(x)
+    out.float().pow(2).mean().backward()
+    dw_fast = m.weight.grad.clone()
+    m.weight.grad = None
+    ref_out = torch.nn.functional.conv2d(
+        x.to(torch.bfloat16), m.weight.to(torch.bfloat16), None,
+        stride=2, padding=3)
+    ref_out.float().pow(2).mean().backward()
+    dw_ref = m.weight.grad
+    assert torch.allclose(out.float(), ref_out.float(), atol=0.1,
+                          rtol=2e-2)
+    rel = (dw_fast - dw_ref).abs().max() / dw_ref.abs().max()
+    assert rel.item() < 3e-2, rel.item()
