@@ -28,6 +28,7 @@ sources = [os.path.join(CSRC, f) for f in (
     "conv_dgrad.hip",
     "conv_bwd.hip",
     "conv_wgrad.hip",
+    "stem_conv.hip",
     "xnor_conv.hip",
     "kurtosis.hip",
     "kd.hip",
