@@ -50,13 +50,10 @@ class StemConv7x7(nn.Conv2d):
                 and x.dim() == 4 and x.size(1) == 3
                 and x.size(2) % 2 == 0 and x.size(3) % 2 == 0
                 and not x.requires_grad
-                and x.dtype in (torch.float32, torch.bfloat16))
+                # the kernel is bf16: only take fp32 inputs when autocast
+                # would have made the stock conv bf16 anyway
+                and (x.dtype == torch.bfloat16
+                     or torch.is_autocast_enabled()))
         if fast:
-            out = _StemConvFn.apply(x, self.weight)
-            # autocast parity: the stock conv would emit bf16 under
-            # autocast and fp32 otherwise; our kernel is bf16-native
-            if (not torch.is_autocast_enabled()
-                    and x.dtype == torch.float32):
-                out = out.float()
-            return out
+            return _StemConvFn.apply(x, self.weight)
         return super().forward(x)
