@@ -28,13 +28,16 @@
 // popcount work.  Word->(tap, offset) address tables are precomputed
 // per block (no integer div/mod in the staging loop).
 #include "common.h"
+#include <cstdlib>
 
-#define TILE_M 128
 #define TILE_K 64
 #define CHUNK 8
-#define A_PER_THREAD 4  // CHUNK*TILE_M / 256
 #define W_PER_THREAD 2  // CHUNK*TILE_K / 256
 #define MAX_WORDS 160   // KH*KW*CW <= 9*16 (C<=512, 3x3); guarded on host
+// TM (tile rows): 128 default; 256 for small-WORDS layers (C <= 128),
+// where the 2-3 chunk main loop is too short to amortize the per-block
+// metadata build and epilogue — doubling rows doubles the popcount work
+// per staged weight word and halves the per-row overheads.
 
 struct XnorConvParams {
   int N, H, W, C, K, KH, KW, stride, pad, Ho, Wo, CW;
@@ -46,7 +49,7 @@ struct XnorConvParams {
 // s1/s2 (nullable): per-out-channel sum / sum-of-squares of the STORED
 // (dtype-rounded) outputs, accumulated in the epilogue — feeds the fused
 // BN directly so BN never re-reads the conv output for its stats pass.
-template <typename TO, bool STATS>
+template <typename TO, bool STATS, int TM>
 __global__ __launch_bounds__(256) void xnor_conv_kernel(
     const uint32_t* __restrict__ xp, const uint32_t* __restrict__ wp,
     const float* __restrict__ alpha, const float* __restrict__ stab,
@@ -66,19 +69,21 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
   int k_blk = wg / grid_m;        // channel tile
   const int tid = threadIdx.x;
   const int64_t M = (int64_t)p.N * p.Ho * p.Wo;
-  const int64_t m0 = (int64_t)m_blk * TILE_M;
+  const int64_t m0 = (int64_t)m_blk * TM;
   const int k0_blk = k_blk * TILE_K;
 
-  __shared__ uint32_t a_lds[2][CHUNK][TILE_M];
+  constexpr int APT = CHUNK * TM / 256;   // a-words staged per thread
+  constexpr int RPT = TM / 16;            // rows per thread tile
+  __shared__ uint32_t a_lds[2][CHUNK][TM];
   __shared__ uint32_t w_lds[2][CHUNK][TILE_K];
-  __shared__ int row_basecw[TILE_M];  // pixel index of tap (0,0), x CW
-  __shared__ unsigned short row_inv[TILE_M];  // invalid-tap bitmask (T<=9)
+  __shared__ int row_basecw[TM];      // pixel index of tap (0,0), x CW
+  __shared__ unsigned short row_inv[TM];      // invalid-tap bitmask (T<=9)
   __shared__ int off_tab[MAX_WORDS];          // (kh*W+kw)*CW + cw per word
   __shared__ unsigned char tap_tab[MAX_WORDS];
   __shared__ float csum[2][TILE_K];           // per-channel stats partials
 
   // ---- per-block metadata ----
-  for (int r = tid; r < TILE_M; r += blockDim.x) {
+  for (int r = tid; r < TM; r += blockDim.x) {
     int64_t sp = m0 + r;
     if (sp >= M) { row_basecw[r] = 0; row_inv[r] = 0xffff; continue; }
     int n = int(sp / ((int64_t)p.Ho * p.Wo));
@@ -107,18 +112,18 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
     }
   __syncthreads();
 
-  // thread's register tile: rows r0..r0+7, channels kq..kq+3
-  const int r0 = (tid / 16) * 8;
+  // thread's register tile: rows r0..r0+RPT-1, channels kq..kq+3
+  const int r0 = (tid / 16) * RPT;
   const int kq = (tid % 16) * 4;
-  int acc[8][4] = {};
+  int acc[RPT][4] = {};
 
   // staging coordinates (fixed per thread)
-  int a_c[A_PER_THREAD], a_r[A_PER_THREAD];
+  int a_c[APT], a_r[APT];
 #pragma unroll
-  for (int it = 0; it < A_PER_THREAD; ++it) {
+  for (int it = 0; it < APT; ++it) {
     int j = tid + it * 256;
-    a_c[it] = j / TILE_M;
-    a_r[it] = j % TILE_M;
+    a_c[it] = j / TM;
+    a_r[it] = j % TM;
   }
   int w_c[W_PER_THREAD], w_k[W_PER_THREAD];
 #pragma unroll
@@ -129,12 +134,12 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
   }
 
   const int n_chunks = (p.WORDS + CHUNK - 1) / CHUNK;
-  uint32_t av[A_PER_THREAD], wv[W_PER_THREAD];
+  uint32_t av[APT], wv[W_PER_THREAD];
 
 #define STAGE_LOAD(w0)                                                    \
   {                                                                       \
     _Pragma("unroll")                                                     \
-    for (int it = 0; it < A_PER_THREAD; ++it) {                           \
+    for (int it = 0; it < APT; ++it) {                                    \
       int widx = (w0) + a_c[it];                                          \
       int r = a_r[it];                                                    \
       uint32_t v = 0;                                                     \
@@ -156,7 +161,7 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
 #define STAGE_WRITE(buf)                                                  \
   {                                                                       \
     _Pragma("unroll")                                                     \
-    for (int it = 0; it < A_PER_THREAD; ++it)                             \
+    for (int it = 0; it < APT; ++it)                                      \
       a_lds[buf][a_c[it]][a_r[it]] = av[it];                              \
     _Pragma("unroll")                                                     \
     for (int it = 0; it < W_PER_THREAD; ++it)                             \
@@ -173,12 +178,13 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
     if (more) STAGE_LOAD((ch + 1) * CHUNK);          // issue early
 #pragma unroll
     for (int c = 0; c < CHUNK; ++c) {
-      uint32_t a8[8], b4[4];
-      *(uint4*)&a8[0] = *(const uint4*)&a_lds[buf][c][r0];
-      *(uint4*)&a8[4] = *(const uint4*)&a_lds[buf][c][r0 + 4];
+      uint32_t a8[RPT], b4[4];
+#pragma unroll
+      for (int q = 0; q < RPT; q += 4)
+        *(uint4*)&a8[q] = *(const uint4*)&a_lds[buf][c][r0 + q];
       *(uint4*)b4 = *(const uint4*)&w_lds[buf][c][kq];
 #pragma unroll
-      for (int i = 0; i < 8; ++i)
+      for (int i = 0; i < RPT; ++i)
 #pragma unroll
         for (int j = 0; j < 4; ++j)
           acc[i][j] += __popc(a8[i] ^ b4[j]);
@@ -198,7 +204,7 @@ __global__ __launch_bounds__(256) void xnor_conv_kernel(
   }
   float st1[STATS ? 4 : 1] = {}, st2[STATS ? 4 : 1] = {};
 #pragma unroll
-  for (int i = 0; i < 8; ++i) {
+  for (int i = 0; i < RPT; ++i) {
     int64_t sp = m0 + r0 + i;
     if (sp >= M) continue;
     unsigned inv = row_inv[r0 + i];
@@ -286,26 +292,36 @@ extern "C" void bdbnn_xnor_conv_fwd(
   int G = (32 * p.CW - C) * p.T;
   p.base = -2 * G - C * p.T;
   int64_t M = (int64_t)N * Ho * Wo;
-  int grid_m = int((M + TILE_M - 1) / TILE_M);
+  // tall tiles for short-K layers (C <= 128 at 3x3): the 2-5 chunk main
+  // loop can't amortize per-block metadata/epilogue at TM=128 — but the
+  // TM=256 variant costs ~100 VGPR of occupancy, so it must WIN a
+  // measurement to be default (BDBNN_XNOR_TM={128,256,0=auto} for A/B)
+  static const int tm_env = [] {
+    const char* e = getenv("BDBNN_XNOR_TM");
+    return e ? atoi(e) : 0;
+  }();
+  int TM = (p.WORDS <= 36) ? 256 : 128;
+  if (tm_env == 128 || tm_env == 256) TM = tm_env;
+  int grid_m = int((M + TM - 1) / TM);
   int grid_k = (K + TILE_K - 1) / TILE_K;
   dim3 grid(grid_m * grid_k);
   if (s1 != nullptr) {
     hipMemsetAsync(s1, 0, sizeof(float) * 32 * K, stream);
     hipMemsetAsync(s2, 0, sizeof(float) * 32 * K, stream);
   }
+#define XLAUNCH(TO, ST, TMV)                                              \
+  xnor_conv_kernel<TO, ST, TMV><<<grid, 256, 0, stream>>>(                \
+      xp, wp, alpha, stab, (TO*)out, s1, s2, p, grid_m)
   if (s1 != nullptr) {
-    if (out_bf16)
-      xnor_conv_kernel<uint16_t, true><<<grid, 256, 0, stream>>>(
-          xp, wp, alpha, stab, (uint16_t*)out, s1, s2, p, grid_m);
-    else
-      xnor_conv_kernel<float, true><<<grid, 256, 0, stream>>>(
-          xp, wp, alpha, stab, (float*)out, s1, s2, p, grid_m);
+    if (out_bf16) { if (TM == 256) XLAUNCH(uint16_t, true, 256);
+                    else XLAUNCH(uint16_t, true, 128); }
+    else          { if (TM == 256) XLAUNCH(float, true, 256);
+                    else XLAUNCH(float, true, 128); }
   } else {
-    if (out_bf16)
-      xnor_conv_kernel<uint16_t, false><<<grid, 256, 0, stream>>>(
-          xp, wp, alpha, stab, (uint16_t*)out, s1, s2, p, grid_m);
-    else
-      xnor_conv_kernel<float, false><<<grid, 256, 0, stream>>>(
-          xp, wp, alpha, stab, (float*)out, s1, s2, p, grid_m);
+    if (out_bf16) { if (TM == 256) XLAUNCH(uint16_t, false, 256);
+                    else XLAUNCH(uint16_t, false, 128); }
+    else          { if (TM == 256) XLAUNCH(float, false, 256);
+                    else XLAUNCH(float, false, 128); }
   }
+#undef XLAUNCH
 }
