@@ -52,6 +52,16 @@ def test_bench_torchrun_4rank_gloo():
 
 
 @pytest.mark.timeout(600)
+def test_bench_torchrun_8rank_gloo():
+    # same world size the driver's 8-GPU SCALE run uses
+    res = _run_bench_dist(8)
+    assert res["n_gpus"] == 8
+    assert res["config"]["parallelism"] == "dp8"
+    assert res["config"]["global_batch"] == 32
+    assert res["value"] > 0
+
+
+@pytest.mark.timeout(600)
 def test_bench_torchrun_2rank_gloo_no_kurt():
     res = _run_bench_dist(2, ("--no-kurt",))
     assert res["n_gpus"] == 2
