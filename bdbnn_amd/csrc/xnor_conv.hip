@@ -292,16 +292,17 @@ extern "C" void bdbnn_xnor_conv_fwd(
   int G = (32 * p.CW - C) * p.T;
   p.base = -2 * G - C * p.T;
   int64_t M = (int64_t)N * Ho * Wo;
-  // tall tiles for short-K layers (C <= 128 at 3x3): the 2-5 chunk main
-  // loop can't amortize per-block metadata/epilogue at TM=128 — but the
-  // TM=256 variant costs ~100 VGPR of occupancy, so it must WIN a
-  // measurement to be default (BDBNN_XNOR_TM={128,256,0=auto} for A/B)
+  // TM=256 ("tall tiles amortize the short-K epilogue") was MEASURED
+  // WORSE everywhere it applied (C=64: 171 vs 201 TbinMAC/s, C=128:
+  // 200 vs 256 — gpurun kernel_bench A/B): the ~100 extra VGPR cost
+  // more occupancy than the longer main loop saved.  Default stays 128;
+  // the template + env override remain for re-testing on new shapes
+  // (BDBNN_XNOR_TM={128,256}).
   static const int tm_env = [] {
     const char* e = getenv("BDBNN_XNOR_TM");
     return e ? atoi(e) : 0;
   }();
-  int TM = (p.WORDS <= 36) ? 256 : 128;
-  if (tm_env == 128 || tm_env == 256) TM = tm_env;
+  int TM = (tm_env == 256) ? 256 : 128;
   int grid_m = int((M + TM - 1) / TM);
   int grid_k = (K + TILE_K - 1) / TILE_K;
   dim3 grid(grid_m * grid_k);

@@ -13,10 +13,19 @@ The stem input never requires grad in this framework, so backward only
 produces the weight gradient (loud error otherwise).
 """
 
+import os
+
 import torch
 import torch.nn as nn
 
 from .. import _C
+
+# wrw backend: the MFMA pixel-split wrw kernel LOSES to MIOpen at the
+# flagship batch (6.06 vs 2.70 ms at b2048, gpurun stem_bench A/B) —
+# its per-128-pixel-tile pT im2col rebuild is LDS-bandwidth bound — so
+# the weight gradient defaults to MIOpen while the forward (2x faster
+# than MIOpen) stays on the owned kernel.  BDBNN_STEM_WRW=mfma for A/B.
+_WRW_MFMA = os.environ.get("BDBNN_STEM_WRW", "miopen") == "mfma"
 
 
 class _StemConvFn(torch.autograd.Function):
@@ -28,14 +37,28 @@ class _StemConvFn(torch.autograd.Function):
             raise RuntimeError("StemConv7x7: input gradients unsupported "
                                "(the stem input is the data batch)")
         if ctx.needs_input_grad[1]:
-            ctx.save_for_backward(x4)
+            if _WRW_MFMA:
+                ctx.save_for_backward(x4)
+            else:
+                xb = x if x.dtype == torch.bfloat16 else x.to(torch.bfloat16)
+                ctx.save_for_backward(
+                    xb.contiguous(memory_format=torch.channels_last))
         return out
 
     @staticmethod
     def backward(ctx, gy):
-        (x4,) = ctx.saved_tensors
-        nat = _C.native_required()
-        dw = nat.stem_conv_wrw(x4, gy)
+        (saved,) = ctx.saved_tensors
+        if _WRW_MFMA:
+            nat = _C.native_required()
+            dw = nat.stem_conv_wrw(saved, gy)
+        else:
+            gb = gy if gy.dtype == torch.bfloat16 else gy.to(torch.bfloat16)
+            dw = torch.ops.aten.convolution_backward(
+                gb.contiguous(memory_format=torch.channels_last), saved,
+                torch.empty(64, 3, 7, 7, device=saved.device,
+                            dtype=torch.bfloat16), None,
+                [2, 2], [3, 3], [1, 1], False, [0, 0], 1,
+                [False, True, False])[1].to(torch.float32)
         return None, dw
 
 

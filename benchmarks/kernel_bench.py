@@ -85,12 +85,12 @@ def main():
     rm = torch.zeros(C, device="cuda")
     rv = torch.ones(C, device="cuda")
     out = nat.bn_act_fwd_train(xcl, gcl, gamma, beta, a, rm, rv, 0.1,
-                               1e-5, 1, None, None)
-    o, z, mean, invstd = out
+                               1e-5, 1, None, None, False)
+    o, z, mean, invstd = out[:4]
     row("bn stats+finalize (1R)",
         timeit(lambda: nat.bn_act_fwd_train(xcl, None, gamma, beta, None,
                                             rm, rv, 0.1, 1e-5, 0, None,
-                                            None)), 3)
+                                            None, False)), 3)
     row("bn_act_bwd (6R2W)",
         timeit(lambda: nat.bn_act_bwd(gcl, z, xcl, mean, invstd, gamma, a,
                                       1, True)), 8)
