@@ -1,6 +1,6 @@
 #!/usr/bin/env bash
 # Round-end GPU validation bundle (run on the box via gpurun):
-#   pytest -m gpu, XNOR TM A/B, stem microbench, e2e bench, rocprof digest.
+#   pytest -m gpu, kernel microbench, stem microbench, e2e bench, rocprof digest.
 set -x
 cd /root/repo
 mkdir -p gpurun_out
@@ -8,8 +8,7 @@ export TMPDIR=/tmp
 timeout 300 python -m pytest tests -m gpu -q > gpurun_out/pytest_gpu.log 2>&1
 echo "pytest_rc=$?" >> gpurun_out/pytest_gpu.log
 tail -3 gpurun_out/pytest_gpu.log
-timeout 200 python benchmarks/kernel_bench.py > gpurun_out/kernel_bench_tm_auto.log 2>&1
-BDBNN_XNOR_TM=128 timeout 200 python benchmarks/kernel_bench.py > gpurun_out/kernel_bench_tm128.log 2>&1
+timeout 200 python benchmarks/kernel_bench.py > gpurun_out/kernel_bench.log 2>&1
 timeout 200 python benchmarks/stem_bench.py 2048 > gpurun_out/stem_bench_2048.log 2>&1
 timeout 420 python bench.py --steps 10 --warmup 3 > gpurun_out/bench_b2048.json 2>&1
 tail -1 gpurun_out/bench_b2048.json
