@@ -55,7 +55,7 @@ void bdbnn_ce_bwd(const void*, const int64_t*, const float*, void*, float,
 void bdbnn_conv_dgrad(const void*, const uint32_t*, const float*, void*,
                       int, int, int, int, int, int, hipStream_t);
 int bdbnn_conv_dgrad2(const void*, const void*, const uint32_t*, void*,
-                      int, int, int, int, int, hipStream_t);
+                      const void*, int, int, int, int, int, hipStream_t);
 void bdbnn_dgrad_wdec(const uint32_t*, const float*, void*, int, int,
                       hipStream_t);
 int bdbnn_conv_wgrad2(const void*, const uint64_t*, float*, int, int, int,
@@ -722,7 +722,8 @@ static bool dgrad2_ok(int H, int W, int C, int K) {
 }
 
 at::Tensor conv_dgrad2(const at::Tensor& g, const at::Tensor& wd,
-                       const at::Tensor& mp, int64_t C) {
+                       const at::Tensor& mp, int64_t C,
+                       const c10::optional<at::Tensor>& acc) {
   TORCH_CHECK(g.is_cuda() && g.dim() == 4 &&
                   g.scalar_type() == at::kBFloat16 &&
                   g.is_contiguous(at::MemoryFormat::ChannelsLast),
@@ -736,11 +737,23 @@ at::Tensor conv_dgrad2(const at::Tensor& g, const at::Tensor& wd,
                   mp.numel() == (int64_t)N * H * W * (C / 32),
               "conv_dgrad2: mask bitplane [...P...][C/32]");
   TORCH_CHECK(dgrad2_ok(H, W, (int)C, K), "conv_dgrad2: unsupported shape");
+  const void* accp = nullptr;
+  at::Tensor ac;
+  if (acc.has_value()) {
+    ac = acc->scalar_type() == at::kBFloat16
+             ? acc->contiguous(at::MemoryFormat::ChannelsLast)
+             : acc->to(at::kBFloat16)
+                   .contiguous(at::MemoryFormat::ChannelsLast);
+    TORCH_CHECK(ac.dim() == 4 && ac.size(0) == N && ac.size(1) == C &&
+                    ac.size(2) == H && ac.size(3) == W,
+                "conv_dgrad2: acc shape must match dx (N,C,H,W)");
+    accp = ac.data_ptr();
+  }
   auto dx = at::empty({N, C, H, W}, g.options(),
                       at::MemoryFormat::ChannelsLast);
   int rc = bdbnn_conv_dgrad2(g.data_ptr(), wd.data_ptr(),
                              (const uint32_t*)mp.data_ptr<int>(),
-                             dx.data_ptr(), N, H, W, (int)C, K,
+                             dx.data_ptr(), accp, N, H, W, (int)C, K,
                              cur_stream());
   TORCH_CHECK(rc == 0, "conv_dgrad2: launch rejected the shape");
   return dx;
@@ -986,7 +999,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_act_eval", &bn_act_eval, "fused BN(+add)(+act) eval forward");
   m.def("conv_dgrad2", &conv_dgrad2,
         "MFMA bf16 dgrad v2: halo-staged 9-tap implicit GEMM with fused "
-        "clip-STE mask (3x3/s1/p1)");
+        "clip-STE mask (3x3/s1/p1); optional acc adds a same-shape "
+        "skip-gradient tensor in the epilogue",
+        py::arg("g"), py::arg("wd"), py::arg("mp"), py::arg("C"),
+        py::arg("acc") = py::none());
   m.def("dgrad_weight_decode", &dgrad_weight_decode,
         "packed bits -> mirrored transposed +-alpha bf16 [9][C][K]");
   m.def("dgrad2_supported",

@@ -58,10 +58,14 @@ struct Dgrad2Params {
 // naive one-tile-per-block version (measured 8x off the MFMA floor at
 // 1 block/CU occupancy) disappears, and consecutive tiles' overlapping
 // halo rows stay L2/L1-warm on the same CU.
-template <int WP, int RB, int GB>
+// ACC: fuse the residual-skip gradient (same-shape bf16 NHWC tensor)
+// into the epilogue store — dx = mask*acc + skip_grad — replacing the
+// separate autograd accumulation pass (one full read+write of dx).
+template <int WP, int RB, int GB, bool ACC>
 __global__ __launch_bounds__(512, 2) void conv_dgrad2_kernel(
     const __bf16* __restrict__ g, const __bf16* __restrict__ wd,
     const uint32_t* __restrict__ mp, __bf16* __restrict__ dx,
+    const __bf16* __restrict__ accp,
     Dgrad2Params p, int grid_m, int nb_m, int tiles_per_block) {
   constexpr int WH = WP + 2;                  // halo row width
   constexpr int NHE = GB * (RB + 2) * WH;     // halo entries (128 B each)
@@ -211,6 +215,9 @@ __global__ __launch_bounds__(512, 2) void conv_dgrad2_kernel(
         if (pix < 0) continue;                                            \
         uint32_t word = mp[(int64_t)pix * p.CW + cw_word];                \
         float v = (word >> cbit) & 1 ? acc[reg] : 0.f;                    \
+        if (ACC)                                                          \
+          v += bf16_to_f32(                                               \
+              *(const uint16_t*)(accp + (int64_t)pix * p.C + ccol));      \
         uint16_t h = f32_to_bf16(v);                                      \
         *(uint16_t*)(dx + (int64_t)pix * p.C + ccol) = h;                 \
       }                                                                   \
@@ -294,7 +301,8 @@ __global__ __launch_bounds__(512, 2) void conv_dgrad2_kernel(
 // ---------------- host launcher ----------------
 
 extern "C" int bdbnn_conv_dgrad2(const void* g, const void* wd,
-                                 const uint32_t* mp, void* dx, int N, int H,
+                                 const uint32_t* mp, void* dx,
+                                 const void* accp, int N, int H,
                                  int W, int C, int K, hipStream_t stream) {
   if (C % 64 || K % 64 || W > 64) return -1;
   Dgrad2Params p;
@@ -313,9 +321,14 @@ extern "C" int bdbnn_conv_dgrad2(const void* g, const void* wd,
     int tpb = (grid_m + nb_m - 1) / nb_m;                                 \
     nb_m = (grid_m + tpb - 1) / tpb;                                      \
     dim3 grid(nb_m * n_ctile);                                            \
-    conv_dgrad2_kernel<WPV, RBV, GBV><<<grid, 512, 0, stream>>>(          \
-        (const __bf16*)g, (const __bf16*)wd, mp, (__bf16*)dx, p, grid_m,  \
-        nb_m, tpb);                                                       \
+    if (accp)                                                             \
+      conv_dgrad2_kernel<WPV, RBV, GBV, true><<<grid, 512, 0, stream>>>(  \
+          (const __bf16*)g, (const __bf16*)wd, mp, (__bf16*)dx,           \
+          (const __bf16*)accp, p, grid_m, nb_m, tpb);                     \
+    else                                                                  \
+      conv_dgrad2_kernel<WPV, RBV, GBV, false><<<grid, 512, 0, stream>>>( \
+          (const __bf16*)g, (const __bf16*)wd, mp, (__bf16*)dx, nullptr,  \
+          p, grid_m, nb_m, tpb);                                          \
     return 0;                                                             \
   }
   if (W <= 8) {

@@ -17,6 +17,8 @@ matching the 19-entry per-layer kurtosis target lists
 (ref:train.py:467-470).
 """
 
+import os
+
 import torch
 import torch.nn as nn
 
@@ -26,6 +28,12 @@ from ..ops.binarize import LearnableBias
 from ..ops.activations import ChannelPReLU
 from ..ops.bn_act import fused_bn_act
 from ..ops.pool import FusedMaxPool2d
+
+
+# defer each bn's residual-skip gradient into the paired conv's dgrad
+# epilogue (kills autograd's separate dx+dskip accumulation pass,
+# ~3.3 ms/step at b2048); see ops/bn_act.py / ops/binary_conv.py
+_FUSE_SKIP_GRAD = os.environ.get("BDBNN_FUSE_SKIP_GRAD", "1") == "1"
 
 
 class FusedDownsample(nn.Sequential):
@@ -110,21 +118,32 @@ class BiBasicBlock(nn.Module):
         pk_in = None
         if isinstance(x, tuple):
             x, pk_in = x
+        # Each bn's skip tensor is also the paired conv's input, so
+        # autograd would sum the two gradient paths in a separate
+        # full-tensor add; a shared per-call cell defers the bn's dskip
+        # into the conv's dgrad epilogue instead (BDBNN_FUSE_SKIP_GRAD=0
+        # restores plain autograd accumulation for A/B).
+        defer = _FUSE_SKIP_GRAD and self.training and x.is_cuda
         if self.downsample is not None:
             identity = self.downsample(x, prepack=pk_in)
+            cell1 = None  # x's grad paths are conv1 + downsample, not bn1
         else:
             identity = x
-        o1, st1 = self.conv1.forward_with_stats(x, prepack=pk_in)
+            cell1 = {} if defer else None
+        o1, st1 = self.conv1.forward_with_stats(x, prepack=pk_in,
+                                                skip_cell=cell1)
         # only a plain ChannelPReLU tail writes the exact conv2 input in
         # its epilogue (RPReLU's shifts run after, so no pack there)
         fuse1 = isinstance(self.act1, ChannelPReLU)
         r1 = fused_bn_act(o1, self.bn1, self.act1, skip=identity, stats=st1,
-                          pack=fuse1)
+                          pack=fuse1, defer_skip_cell=cell1)
         out, pk1 = r1 if fuse1 else (r1, None)
-        o2, st2 = self.conv2.forward_with_stats(out, prepack=pk1)
+        cell2 = {} if defer else None
+        o2, st2 = self.conv2.forward_with_stats(out, prepack=pk1,
+                                                skip_cell=cell2)
         fuse2 = isinstance(self.act2, ChannelPReLU)
         r2 = fused_bn_act(o2, self.bn2, self.act2, skip=out, stats=st2,
-                          pack=fuse2)
+                          pack=fuse2, defer_skip_cell=cell2)
         out, pk2 = r2 if fuse2 else (r2, None)
         return (out, pk2) if pk2 is not None else out
 

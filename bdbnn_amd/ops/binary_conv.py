@@ -63,13 +63,18 @@ class BinaryConvFunction(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w, stride, padding, act_mode, t, k,
-                want_stats=False, xp_pre=None, mp_pre=None):
+                want_stats=False, xp_pre=None, mp_pre=None, skip_cell=None):
         ctx.stride = stride
         ctx.padding = padding
         ctx.act_mode = act_mode
         ctx.t = t
         ctx.k = k
         ctx.packed = False
+        # deferred residual-skip gradient: the consuming BN's backward
+        # (which runs first) stashes its dskip here instead of returning
+        # it, and this conv's dgrad adds it in the epilogue — removing
+        # autograd's separate grad-accumulation pass over dx
+        ctx.skip_cell = skip_cell
         if x.is_cuda:
             nat = _C.native_required()
             wp, alpha, stab = nat.weight_pack(w)  # bits, alpha[K], S[K][T]
@@ -122,6 +127,7 @@ class BinaryConvFunction(torch.autograd.Function):
     @staticmethod
     def backward(ctx, g, _gs1=None, _gs2=None):
         stride, padding = ctx.stride, ctx.padding
+        acc = ctx.skip_cell.pop("g", None) if ctx.skip_cell else None
         if ctx.packed:
             w, xp, mp, wp, alpha = ctx.saved_tensors
             nat = _C.native_required()
@@ -136,9 +142,10 @@ class BinaryConvFunction(torch.autograd.Function):
                     and stride == 1 and padding == 1 and kh == 3
                     and nat.dgrad2_supported(H, W, C, K))
             if use2:
-                # dgrad: halo implicit GEMM, clip-STE mask fused
+                # dgrad: halo implicit GEMM, clip-STE mask (+ deferred
+                # skip grad, if any) fused into the epilogue
                 wd = nat.dgrad_weight_decode(wp, alpha, C)
-                dx = nat.conv_dgrad2(g, wd, mp, C)
+                dx = nat.conv_dgrad2(g, wd, mp, C, acc)
                 # wgrad: all-9-tap block GEMM straight from the sign
                 # BITS (the dense +-1 activation tensor is never
                 # materialized), then transpose + |w|<=1 mask in one pass
@@ -146,7 +153,7 @@ class BinaryConvFunction(torch.autograd.Function):
                 dwT = nat.conv_wgrad2(g, xcp, C)
                 dw = nat.wgrad_finish(dwT, w.float())
                 return (dx, dw.to(w.dtype), None, None, None, None, None,
-                        None, None, None)
+                        None, None, None, None)
             xb = nat.decode_packed(xp, C, bf16)
             wb = nat.weight_decode(wp, alpha, C, bf16)
             dxb, dwb = torch.ops.aten.convolution_backward(
@@ -155,9 +162,11 @@ class BinaryConvFunction(torch.autograd.Function):
                 1, [True, True, False])[:2]
             dx = nat.mask_mul_packed(dxb, mp, C,
                                      ctx.x_dtype == torch.bfloat16)
+            if acc is not None:
+                dx = dx + acc.to(dx.dtype)
             dw = nat.ste_mask_mul(dwb, w, 0, 0.0, 0.0)
             return (dx, dw.to(w.dtype), None, None, None, None, None,
-                    None, None, None)
+                    None, None, None, None)
         x, w = ctx.saved_tensors
         if x.is_cuda:
             nat = _C.native_required()
@@ -185,8 +194,10 @@ class BinaryConvFunction(torch.autograd.Function):
         else:
             dx = dxb * _act_grad_mask(x, ctx.act_mode, ctx.t, ctx.k)
             dw = dwb * (w.abs() <= 1).to(w.dtype)
+        if acc is not None:
+            dx = dx + acc.to(dx.dtype)
         return (dx, dw.to(w.dtype), None, None, None, None, None, None,
-                None, None)
+                None, None, None)
 
 
 class _HardBinaryConvBase(nn.Module):
@@ -236,16 +247,16 @@ class _HardBinaryConvBase(nn.Module):
             return xp, mp
         return None, None
 
-    def forward(self, x, prepack=None):
+    def forward(self, x, prepack=None, skip_cell=None):
         t = float(self.t) if self.t is not None else None
         k = float(self.k) if self.k is not None else None
         xp, mp = self._check_prepack(x, prepack)
         out, _, _ = BinaryConvFunction.apply(
             x, self.weight, self.stride, self.padding, self.act_mode, t, k,
-            False, xp, mp)
+            False, xp, mp, skip_cell)
         return out
 
-    def forward_with_stats(self, x, prepack=None):
+    def forward_with_stats(self, x, prepack=None, skip_cell=None):
         """(out, (s1, s2)|None): per-out-channel sum/sumsq accumulated in
         the conv epilogue, for the fused BN that consumes the output.
 
@@ -261,7 +272,7 @@ class _HardBinaryConvBase(nn.Module):
         xp, mp = self._check_prepack(x, prepack)
         out, s1, s2 = BinaryConvFunction.apply(
             x, self.weight, self.stride, self.padding, self.act_mode, t, k,
-            want, xp, mp)
+            want, xp, mp, skip_cell)
         return out, ((s1, s2) if want else None)
 
 

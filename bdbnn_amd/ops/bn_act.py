@@ -25,7 +25,7 @@ class _BNActFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, skip, gamma, beta, a, running_mean, running_var,
                 momentum, eps, act_kind, training, s1=None, s2=None,
-                want_pack=False):
+                want_pack=False, defer_cell=None):
         nat = _C.native_required()
         res = nat.bn_act_fwd_train(
             x, skip, gamma, beta, a, running_mean, running_var,
@@ -36,6 +36,11 @@ class _BNActFn(torch.autograd.Function):
         ctx.act_kind = act_kind
         ctx.has_skip = skip is not None
         ctx.has_a = a is not None
+        # deferred skip grad: instead of returning dskip (which autograd
+        # would add to the conv's dx in a separate full-tensor pass), the
+        # backward stashes it in this cell; the conv that produced the
+        # skip tensor's other consumer fuses it into its dgrad epilogue.
+        ctx.defer_cell = defer_cell if skip is not None else None
         if want_pack:
             # next conv's sign/mask bitplanes, packed in the epilogue
             xpk, mpk = res[4], res[5]
@@ -52,15 +57,19 @@ class _BNActFn(torch.autograd.Function):
         dx, dskip, dgamma, dbeta, da = nat.bn_act_bwd(
             dy, z, x, mean, invstd, gamma,
             a if ctx.has_a else None, ctx.act_kind, ctx.has_skip)
-        return (dx,
-                dskip if ctx.has_skip else None,
+        skip_grad = dskip if ctx.has_skip else None
+        if ctx.defer_cell is not None and skip_grad is not None:
+            ctx.defer_cell["g"] = skip_grad
+            skip_grad = None
+        return (dx, skip_grad,
                 dgamma, dbeta,
                 da if ctx.has_a else None,
-                None, None, None, None, None, None, None, None, None)
+                None, None, None, None, None, None, None, None, None,
+                None)
 
 
 def fused_bn_act(x, bn: nn.BatchNorm2d, act=None, skip=None, stats=None,
-                 pack=False):
+                 pack=False, defer_skip_cell=None):
     """BN(x) (+skip) then act.  act: None | ChannelPReLU | 'relu'.
 
     stats: optional (sum, sumsq) per channel of x, pre-accumulated by the
@@ -77,7 +86,8 @@ def fused_bn_act(x, bn: nn.BatchNorm2d, act=None, skip=None, stats=None,
     elif act is None:
         act_kind, a = _ACT_NONE, None
     else:  # generic module: apply unfused after BN+add
-        out = fused_bn_act(x, bn, None, skip, stats=stats)
+        out = fused_bn_act(x, bn, None, skip, stats=stats,
+                           defer_skip_cell=defer_skip_cell)
         out = act(out)
         return (out, None) if pack else out
 
@@ -92,7 +102,7 @@ def fused_bn_act(x, bn: nn.BatchNorm2d, act=None, skip=None, stats=None,
         out, xpk, mpk = _BNActFn.apply(
             x, skip, bn.weight, bn.bias, a, bn.running_mean, bn.running_var,
             bn.momentum if bn.momentum is not None else 0.1, bn.eps,
-            act_kind, True, s1, s2, want_pack)
+            act_kind, True, s1, s2, want_pack, defer_skip_cell)
         if pack:
             return out, ((xpk, mpk) if want_pack else None)
         return out

@@ -664,6 +664,61 @@ def test_conv_dgrad2_matches_reference(N, C, H, K):
     assert torch.allclose(dx.float(), _cl(ref), atol=0.5, rtol=2e-2), err
 
 
+@pytest.mark.parametrize("N,C,H,K", [(2, 64, 56, 64), (4, 512, 7, 512)])
+def test_conv_dgrad2_acc_fuses_skip_grad(N, C, H, K):
+    """The optional acc operand must equal dgrad2 + a separate add of
+    the (unmasked) skip gradient."""
+    torch.manual_seed(33)
+    nat = _nat()
+    g = _cl(torch.randn(N, K, H, H, device="cuda", dtype=torch.bfloat16))
+    w = torch.randn(K, C, 3, 3, device="cuda")
+    x = torch.randn(N, C, H, H, device="cuda")
+    skip_g = _cl(torch.randn(N, C, H, H, device="cuda",
+                             dtype=torch.bfloat16))
+    wp, alpha, _ = nat.weight_pack(w)
+    _, mp = nat.sign_mask_pack_nhwc(_cl(x))
+    wd = nat.dgrad_weight_decode(wp, alpha, C)
+    base = nat.conv_dgrad2(g, wd, mp, C)
+    fused = nat.conv_dgrad2(g, wd, mp, C, skip_g)
+    want = (base.float() + skip_g.float()).to(torch.bfloat16)
+    assert torch.allclose(fused.float(), want.float(), atol=2e-2,
+                          rtol=1e-2), (fused.float() -
+                                       want.float()).abs().max().item()
+
+
+def test_bi_block_defer_skip_grad_matches_plain():
+    """BiBasicBlock grads with the deferred-skip-grad path (default on)
+    vs plain autograd accumulation, same weights/inputs."""
+    from bdbnn_amd.models import resnet_common as rc
+    torch.manual_seed(34)
+    blk = rc.BiBasicBlock(64, 64).cuda().to(
+        memory_format=torch.channels_last)
+    x0 = torch.randn(2, 64, 14, 14, device="cuda")
+    gout = torch.randn(2, 64, 14, 14, device="cuda")
+
+    def run(defer):
+        old = rc._FUSE_SKIP_GRAD
+        rc._FUSE_SKIP_GRAD = defer
+        try:
+            blk.zero_grad(set_to_none=True)
+            x = _cl(x0.clone()).requires_grad_(True)
+            out = blk(x)
+            if isinstance(out, tuple):
+                out = out[0]
+            out.backward(_cl(gout))
+            return ([p.grad.clone() for p in blk.parameters()],
+                    x.grad.clone())
+        finally:
+            rc._FUSE_SKIP_GRAD = old
+
+    pg_d, xg_d = run(True)
+    pg_p, xg_p = run(False)
+    assert torch.allclose(xg_d, xg_p, atol=1e-4, rtol=1e-3), \
+        (xg_d - xg_p).abs().max().item()
+    for a, b in zip(pg_d, pg_p):
+        assert torch.allclose(a, b, atol=1e-4, rtol=1e-3)
+
+
 def test_dgrad_weight_decode_values():
     torch.manual_seed(32)
     nat = _nat()
