@@ -208,16 +208,28 @@ __global__ __launch_bounds__(512, 2) void conv_dgrad2_kernel(
     for (int half = 0; half < 2; ++half) {                                \
       const f32x16& acc = half ? acc1 : acc0;                             \
       const int mbase = wm0 + half * 32 + 4 * lk8;                        \
+      int pixr[16];                                                       \
+      uint32_t mpw[16];                                                   \
+      uint16_t av[16];                                                    \
+      _Pragma("unroll")                                                   \
+      for (int reg = 0; reg < 16; ++reg)                                  \
+        pixr[reg] = tab_out[tb][mbase + (reg & 3) + 8 * (reg >> 2)];      \
+      /* batch-issue the mask/skip-grad loads (one waitcnt for all,   */ \
+      /* instead of a load-use pair per element that serializes the   */ \
+      /* whole epilogue behind HBM latency once per tile).  pix<0     */ \
+      /* lanes read a safe dummy address and are masked below.        */ \
       _Pragma("unroll")                                                   \
       for (int reg = 0; reg < 16; ++reg) {                                \
-        int m = mbase + (reg & 3) + 8 * (reg >> 2);                       \
-        int pix = tab_out[tb][m];                                         \
+        int64_t pz = pixr[reg] < 0 ? 0 : (int64_t)pixr[reg];              \
+        mpw[reg] = mp[pz * p.CW + cw_word];                               \
+        if (ACC) av[reg] = *(const uint16_t*)(accp + pz * p.C + ccol);    \
+      }                                                                   \
+      _Pragma("unroll")                                                   \
+      for (int reg = 0; reg < 16; ++reg) {                                \
+        int pix = pixr[reg];                                              \
         if (pix < 0) continue;                                            \
-        uint32_t word = mp[(int64_t)pix * p.CW + cw_word];                \
-        float v = (word >> cbit) & 1 ? acc[reg] : 0.f;                    \
-        if (ACC)                                                          \
-          v += bf16_to_f32(                                               \
-              *(const uint16_t*)(accp + (int64_t)pix * p.C + ccol));      \
+        float v = (mpw[reg] >> cbit) & 1 ? acc[reg] : 0.f;                \
+        if (ACC) v += bf16_to_f32(av[reg]);                               \
         uint16_t h = f32_to_bf16(v);                                      \
         *(uint16_t*)(dx + (int64_t)pix * p.C + ccol) = h;                 \
       }                                                                   \
