@@ -106,6 +106,9 @@ def build_parser():
                    help="bf16 autocast for the dense (stem/head/backward) path")
     p.add_argument("--synthetic-data", action="store_true",
                    help="synthetic dataset of the selected shape (offline)")
+    p.add_argument("--synthetic-train-len", default=100000, type=int,
+                   help="length of the synthetic imagenet train split "
+                        "(short epochs for smoke/parity runs)")
     p.add_argument("--auto-resume", action="store_true",
                    help="resume from <log_path>/checkpoint.pth.tar if present "
                         "(crash recovery; the reference requires manual --resume)")
@@ -154,6 +157,8 @@ def make_loaders(args, distributed):
         per_rank_bs = max(1, args.batch_size // dist.get_world_size())
     kw = dict(batch_size=per_rank_bs, data_path=args.data,
               workers=args.workers, synthetic=args.synthetic_data)
+    if args.dataset == "imagenet":
+        kw["synthetic_len"] = args.synthetic_train_len
     fn = {"cifar10": dataloader_cifar10, "cifar100": dataloader_cifar100,
           "imagenet": dataloader_imagenet}[args.dataset]
     train_loader = fn("train", distributed=distributed, **kw)
