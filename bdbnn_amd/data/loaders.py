@@ -117,9 +117,15 @@ class CIFARBase(Dataset):
     mean = _CIFAR_MEAN
     std = _CIFAR_STD
 
+    # archive the reference's torchvision download=True would fetch
+    # (ref:loader.py:23-26); this offline image cannot download, but a
+    # user-provided tarball in the dataset root is auto-extracted
+    _archive = None
+
     def __init__(self, root, train=True, augment=True):
         self.train = train
         self.augment = augment and train
+        self._maybe_extract(root)
         data, labels = [], []
         for fname in self._files(train):
             path = os.path.join(root, self._subdir(), fname)
@@ -129,6 +135,18 @@ class CIFARBase(Dataset):
             labels.extend(d.get("labels", d.get("fine_labels")))
         self.data = np.concatenate(data).reshape(-1, 3, 32, 32)
         self.labels = np.asarray(labels, dtype=np.int64)
+
+    def _maybe_extract(self, root):
+        """Extract <root>/<archive>.tar.gz if the batch dir is absent —
+        the no-network stand-in for the reference's download=True."""
+        sub = os.path.join(root, self._subdir())
+        if os.path.isdir(sub) or not self._archive:
+            return
+        arc = os.path.join(root, self._archive)
+        if os.path.isfile(arc):
+            import tarfile
+            with tarfile.open(arc, "r:gz") as tf:
+                tf.extractall(root)
 
     def __len__(self):
         return len(self.labels)
@@ -147,6 +165,8 @@ class CIFARBase(Dataset):
 
 
 class CIFAR10Dataset(CIFARBase):
+    _archive = "cifar-10-python.tar.gz"
+
     def _subdir(self):
         return "cifar-10-batches-py"
 
@@ -156,6 +176,8 @@ class CIFAR10Dataset(CIFARBase):
 
 
 class CIFAR100Dataset(CIFARBase):
+    _archive = "cifar-100-python.tar.gz"
+
     def _subdir(self):
         return "cifar-100-python"
 
@@ -258,7 +280,7 @@ def _make_loader(dataset, batch_size, shuffle, workers, distributed,
                       persistent_workers=workers > 0)
 
 
-def _resolve_dataset_dir(data_path, subdir, synthetic, what):
+def _resolve_dataset_dir(data_path, subdir, synthetic, what, archive=None):
     """Fallback policy: synthetic data is used only when asked for
     (``synthetic=True``) or when no path was given at all (then with a
     loud warning).  A missing EXPLICIT path raises — a run pointed at a
@@ -274,6 +296,8 @@ def _resolve_dataset_dir(data_path, subdir, synthetic, what):
         return None
     full = os.path.join(data_path, subdir) if subdir else data_path
     if not os.path.isdir(full):
+        if archive and os.path.isfile(os.path.join(data_path, archive)):
+            return full  # the dataset ctor auto-extracts the tarball
         raise FileNotFoundError(
             f"{what}: dataset directory {full!r} does not exist "
             "(pass --synthetic-data for synthetic random data)")
@@ -284,7 +308,8 @@ def dataloader_cifar10(split="train", batch_size=128, data_path=None,
                        distributed=False, workers=4, synthetic=False):
     train = split == "train"
     found = _resolve_dataset_dir(data_path, "cifar-10-batches-py",
-                                 synthetic, "cifar10")
+                                 synthetic, "cifar10",
+                                 archive="cifar-10-python.tar.gz")
     if found is None:
         ds = SyntheticImageDataset(50000 if train else 10000, (3, 32, 32), 10)
     else:
@@ -296,7 +321,8 @@ def dataloader_cifar100(split="train", batch_size=128, data_path=None,
                         distributed=False, workers=4, synthetic=False):
     train = split == "train"
     found = _resolve_dataset_dir(data_path, "cifar-100-python",
-                                 synthetic, "cifar100")
+                                 synthetic, "cifar100",
+                                 archive="cifar-100-python.tar.gz")
     if found is None:
         ds = SyntheticImageDataset(50000 if train else 10000, (3, 32, 32), 100)
     else:

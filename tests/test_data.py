@@ -56,6 +56,27 @@ def test_dataloader_cifar10_real_files(tmp_path):
     assert x.shape == (8, 3, 32, 32)
 
 
+def test_cifar_tarball_bootstrap(tmp_path):
+    """A cifar-10-python.tar.gz in the dataset root is auto-extracted
+    (the no-network stand-in for the reference's download=True,
+    ref:loader.py:23-26)."""
+    import shutil
+    import tarfile
+    staging = tmp_path / "staging"
+    _write_fake_cifar(str(staging))
+    root = tmp_path / "data"
+    os.makedirs(root)
+    with tarfile.open(root / "cifar-10-python.tar.gz", "w:gz") as tf:
+        tf.add(staging / "cifar-10-batches-py",
+               arcname="cifar-10-batches-py")
+    shutil.rmtree(staging)
+    loader = dataloader_cifar10("train", batch_size=8,
+                                data_path=str(root), workers=0)
+    x, y = next(iter(loader))
+    assert x.shape == (8, 3, 32, 32)
+    assert (root / "cifar-10-batches-py" / "data_batch_1").exists()
+
+
 def test_dataloader_missing_explicit_path_raises(tmp_path):
     # a wrong --data path must not silently 'train' on noise
     import pytest
