@@ -719,6 +719,44 @@ def test_bi_block_defer_skip_grad_matches_plain():
         assert torch.allclose(a, b, atol=1e-4, rtol=1e-3)
 
 
+def test_bi_block_defer_skip_grad_bf16_dgrad2_path():
+    """Same as above but bf16/autocast on a dgrad2-supported shape, so
+    the deferred skip grad rides the fused conv_dgrad2 acc operand."""
+    from bdbnn_amd.models import resnet_common as rc
+    torch.manual_seed(35)
+    blk = rc.BiBasicBlock(64, 64).cuda().to(
+        memory_format=torch.channels_last)
+    x0 = torch.randn(2, 64, 56, 56, device="cuda", dtype=torch.bfloat16)
+    gout = torch.randn(2, 64, 56, 56, device="cuda",
+                       dtype=torch.bfloat16)
+
+    def run(defer):
+        old = rc._FUSE_SKIP_GRAD
+        rc._FUSE_SKIP_GRAD = defer
+        try:
+            blk.zero_grad(set_to_none=True)
+            x = _cl(x0.clone()).requires_grad_(True)
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                out = blk(x)
+            if isinstance(out, tuple):
+                out = out[0]
+            out.backward(_cl(gout))
+            return ([p.grad.clone() for p in blk.parameters()],
+                    x.grad.clone())
+        finally:
+            rc._FUSE_SKIP_GRAD = old
+
+    pg_d, xg_d = run(True)
+    pg_p, xg_p = run(False)
+    # only rounding differs: fused adds in f32 then rounds once; plain
+    # adds two already-rounded bf16 tensors
+    assert torch.allclose(xg_d.float(), xg_p.float(), atol=3e-2,
+                          rtol=2e-2), (xg_d.float() -
+                                       xg_p.float()).abs().max().item()
+    for a, b in zip(pg_d, pg_p):
+        assert torch.allclose(a.float(), b.float(), atol=3e-2, rtol=2e-2)
+
+
 def test_dgrad_weight_decode_values():
     torch.manual_seed(32)
     nat = _nat()
