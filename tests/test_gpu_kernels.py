@@ -749,12 +749,19 @@ def test_bi_block_defer_skip_grad_bf16_dgrad2_path():
     pg_d, xg_d = run(True)
     pg_p, xg_p = run(False)
     # only rounding differs: fused adds in f32 then rounds once; plain
-    # adds two already-rounded bf16 tensors
+    # adds two already-rounded bf16 tensors.  The input grad (the direct
+    # consumer of the fused add) must agree elementwise; param grads sit
+    # behind large bf16 reductions of that rounded tensor where
+    # elementwise cancellation amplifies the rounding delta, so they are
+    # compared in relative norm (a routing bug — lost/duplicated dskip —
+    # would show up as an O(1) relative error; the fp32 variant above
+    # checks exact routing at 1e-4).
     assert torch.allclose(xg_d.float(), xg_p.float(), atol=3e-2,
                           rtol=2e-2), (xg_d.float() -
                                        xg_p.float()).abs().max().item()
     for a, b in zip(pg_d, pg_p):
-        assert torch.allclose(a.float(), b.float(), atol=3e-2, rtol=2e-2)
+        rel = (a.float() - b.float()).norm() / (b.float().norm() + 1e-12)
+        assert rel.item() < 0.05, rel.item()
 
 
 def test_dgrad_weight_decode_values():
