@@ -34,10 +34,11 @@
 #define CHUNK 8
 #define W_PER_THREAD 2  // CHUNK*TILE_K / 256
 #define MAX_WORDS 160   // KH*KW*CW <= 9*16 (C<=512, 3x3); guarded on host
-// TM (tile rows): 128 default; 256 for small-WORDS layers (C <= 128),
-// where the 2-3 chunk main loop is too short to amortize the per-block
-// metadata build and epilogue — doubling rows doubles the popcount work
-// per staged weight word and halves the per-row overheads.
+// TM (tile rows): 128 everywhere.  The TM=256 variant (meant to
+// amortize the short 2-3 chunk main loop of C<=128 layers) measured
+// WORSE on every shape it applied to — its ~100 extra VGPRs cost more
+// occupancy than the amortization saved (profiles/r02_stem_tm_ab.md);
+// it stays compiled for BDBNN_XNOR_TM=256 A/B on future shapes.
 
 struct XnorConvParams {
   int N, H, W, C, K, KH, KW, stride, pad, Ho, Wo, CW;
