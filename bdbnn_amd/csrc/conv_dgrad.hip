@@ -1,4 +1,7 @@
-// EXPERIMENTAL (round-2 seed, default-off): hand-written MFMA bf16 dgrad
+// SUPERSEDED on the hot path by csrc/conv_bwd.hip (dgrad2): kept as an
+// independent rocWMMA implementation — a second oracle the v2 kernel's
+// GPU numerics tests cross-check against, and the minimal readable
+// form of the algorithm.  Hand-written MFMA bf16 dgrad
 // for the 3x3/stride-1/pad-1 binary convs.
 //
 //   dx[n,iy,ix,c] = sum_{kh,kw,k} g[n, iy+1-kh, ix+1-kw, k] * wb[k,c,kh,kw]

@@ -1,4 +1,7 @@
-// EXPERIMENTAL (round-2 seed, default-off): MFMA bf16 wgrad for the
+// SUPERSEDED on the hot path by csrc/conv_bwd.hip (wgrad2): kept as an
+// independent rocWMMA implementation — a second oracle the v2 kernel's
+// GPU numerics tests cross-check against, and the minimal readable
+// form of the algorithm.  MFMA bf16 wgrad for the
 // 3x3/stride-1/pad-1 binary convs.
 //
 //   dW[k,c,kh,kw] = sum_{n,oy,ox} g[n,oy,ox,k] * xb[n,oy-1+kh,ox-1+kw,c]
