@@ -591,6 +591,26 @@ def test_experimental_conv_dgrad_matches_reference():
             (dx.float() - _cl(ref)).abs().max().item()
 
 
+def test_dgrad_v1_v2_cross_check():
+    """The two independent MFMA dgrad implementations (v1 rocWMMA
+    in-kernel decode, v2 halo implicit GEMM) must agree: v2's fused
+    mask == mask_mul(v1)."""
+    torch.manual_seed(25)
+    nat = _nat()
+    N, C, H, K = 2, 64, 56, 64
+    g = _cl(torch.randn(N, K, H, H, device="cuda", dtype=torch.bfloat16))
+    w = torch.randn(K, C, 3, 3, device="cuda")
+    x = torch.randn(N, C, H, H, device="cuda")
+    wp, alpha, _ = nat.weight_pack(w)
+    _, mp = nat.sign_mask_pack_nhwc(_cl(x))
+    wd = nat.dgrad_weight_decode(wp, alpha, C)
+    dx2 = nat.conv_dgrad2(g, wd, mp, C)
+    dx1 = nat.mask_mul_packed(nat.conv_dgrad(g, wp, alpha, C), mp, C, True)
+    assert torch.allclose(dx2.float(), dx1.float(), atol=5e-2,
+                          rtol=2e-2), (dx2.float() -
+                                       dx1.float()).abs().max().item()
+
+
 def test_experimental_conv_wgrad_matches_reference():
     # validated on MI355X (first-shot numerics pass); default-off in the
     # training path until round-2 perf tuning
