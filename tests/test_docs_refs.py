@@ -9,7 +9,7 @@ import subprocess
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 DOCS = ["README.md", "ARCHITECTURE.md", "docs/KERNELS.md",
-        "docs/TUNING.md", "docs/R2_SUMMARY.md"]
+        "docs/TUNING.md", "docs/R2_SUMMARY.md", "docs/SCALING.md"]
 
 
 def _doc_text():
