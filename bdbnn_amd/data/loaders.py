@@ -145,7 +145,17 @@ class CIFARBase(Dataset):
         arc = os.path.join(root, self._archive)
         if os.path.isfile(arc):
             import tarfile
+            rootp = os.path.realpath(root)
             with tarfile.open(arc, "r:gz") as tf:
+                for m in tf.getmembers():
+                    # refuse path traversal / links out of the root
+                    dest = os.path.realpath(os.path.join(rootp, m.name))
+                    if not dest.startswith(rootp + os.sep):
+                        raise RuntimeError(
+                            f"{self._archive}: unsafe member {m.name!r}")
+                    if not (m.isreg() or m.isdir()):
+                        raise RuntimeError(
+                            f"{self._archive}: non-file member {m.name!r}")
                 tf.extractall(root)
 
     def __len__(self):
