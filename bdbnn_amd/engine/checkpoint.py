@@ -3,7 +3,6 @@
 best_acc1, optimizer} saved as checkpoint.pth.tar, best copied to
 model_best.pth.tar."""
 
-import os
 
 import torch
 
